@@ -1,0 +1,202 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: global-shuffle sample fetch through the MI355X-native
+store, side-stream prefetch overlapped with a bf16 train step (BASELINE.json
+configs 2/3/5 rolled into the end-to-end path; the reference publishes no
+numbers -- BASELINE.md).
+
+Per rank (one process per GPU, RCCL over xGMI for N>1):
+  * shard: ``--rows`` rows x ``--dim`` float32 in local HBM (default
+    2 Mi x 128 = 1 GiB; 512 B rows -- the reference's per-get granularity,
+    test/demo.py:45-50)
+  * each step fetches ``--batch`` globally-shuffled rows (fraction (N-1)/N
+    remote over xGMI) with the CDNA4 gather kernel on a side stream, while
+    the main stream runs the bf16 MLP train step (forward, backward,
+    optimizer) on the previous batch -- nothing is skipped in the timed loop.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints ONE JSON line: whole-job samples/sec (aggregate over ranks),
+plus effective gather + remote-read GB/s in ``extra``.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--rows", type=int, default=2 * 1024 * 1024, help="rows per rank")
+    p.add_argument("--dim", type=int, default=128, help="f32 elems per row (128 -> 512 B rows)")
+    p.add_argument("--batch", type=int, default=131072, help="rows fetched per rank per step")
+    p.add_argument("--hidden", type=int, default=1024, help="MLP hidden width")
+    p.add_argument("--mode", choices=["train", "fetch"], default="train",
+                   help="train = fetch overlapped with bf16 MLP step (flagship); "
+                        "fetch = pure gather throughput")
+    p.add_argument("--device", default="cuda")
+    return p.parse_args()
+
+
+class TrainStep:
+    """bf16 MLP forward+backward+SGD consuming a fetched minibatch."""
+
+    def __init__(self, dim: int, hidden: int, device: torch.device, world: int):
+        self.model = torch.nn.Sequential(
+            torch.nn.Linear(dim, hidden),
+            torch.nn.GELU(),
+            torch.nn.Linear(hidden, dim),
+        ).to(device=device, dtype=torch.bfloat16)
+        if world > 1:
+            self.model = torch.nn.parallel.DistributedDataParallel(self.model)
+        self.opt = torch.optim.SGD(self.model.parameters(), lr=1e-3)
+
+    def __call__(self, batch_bf16: torch.Tensor):
+        self.opt.zero_grad(set_to_none=True)
+        out = self.model(batch_bf16)
+        loss = torch.nn.functional.mse_loss(out, batch_bf16)
+        loss.backward()
+        self.opt.step()
+        return loss
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    use_cuda = args.device.startswith("cuda") and torch.cuda.is_available()
+
+    if world > 1:
+        backend = "nccl" if use_cuda else "gloo"
+        dist.init_process_group(backend, rank=rank, world_size=world)
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    from ddstore_amd import DDStore, PrefetchLoader
+
+    store = DDStore(device=device if use_cuda else "cpu")
+    rows, dim, batch = args.rows, args.dim, args.batch
+    shard = torch.randn(rows, dim, dtype=torch.float32,
+                        device=device if use_cuda else "cpu")
+    store.add("bench", shard)
+    del shard
+
+    ntotal = rows * world
+    nsteps_total = args.warmup + args.steps
+    # sampler: per-epoch global permutation, this rank's disjoint slice
+    # (DistributedSampler semantics, reference vae-ddp.py:216)
+    g = torch.Generator().manual_seed(1234)
+    need = nsteps_total * batch
+    order = []
+    got = 0
+    epoch = 0
+    while got < need:
+        perm = torch.randperm(ntotal, generator=g)
+        share = perm[rank::world]  # this rank's slice of the global shuffle
+        order.append(share)
+        got += share.numel()
+        epoch += 1
+    order = torch.cat(order)[:need]
+
+    trainer = TrainStep(dim, args.hidden, device, world) if args.mode == "train" else None
+
+    loader = PrefetchLoader(store, "bench", order, batch, out_dtype=torch.bfloat16,
+                            depth=3, drop_last=True)
+
+    def run_steps(it, n: int):
+        for _ in range(n):
+            b = next(it)
+            if trainer is not None:
+                trainer(b)
+            else:
+                # fetch mode: fold the batch into a scalar so the gather is
+                # consumed on the main stream (not dead code)
+                b.float().sum()
+
+    store.epoch_begin()
+    it = iter(loader)
+    run_steps(it, args.warmup)
+
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    run_steps(it, args.steps)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t1 = time.perf_counter()
+    store.epoch_end()
+
+    elapsed = t1 - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu",
+                         dtype=torch.float64)
+        if dist.get_backend() == "nccl":
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        else:
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_samples = world * args.steps * batch
+    row_bytes = dim * 4
+    sps = n_samples / elapsed
+    gather_gbps = n_samples * row_bytes / elapsed / 1e9
+    remote_gbps = gather_gbps * (world - 1) / world if world > 0 else 0.0
+
+    if rank == 0:
+        result = {
+            "metric": "samples/sec (global-shuffle fetch + bf16 train step)",
+            "value": sps,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": f"store-globalshuffle+bf16-mlp{args.hidden}" if args.mode == "train"
+                         else "store-globalshuffle-fetch",
+                "global_batch": world * batch,
+                "rows_per_rank": rows,
+                "row_bytes": row_bytes,
+                "shard_GiB": rows * row_bytes / 2**30,
+                "parallelism": f"dp{world}",
+                "mode": args.mode,
+            },
+            "extra": {
+                "gather_GBps_aggregate": gather_gbps,
+                "remote_read_GBps_aggregate": remote_gbps,
+                "store_stats_rank0": store.stats().get("bench", {}),
+            },
+        }
+        print(json.dumps(result))
+
+    store.free()
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

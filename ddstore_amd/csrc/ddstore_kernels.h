@@ -1,0 +1,69 @@
+// ddstore_amd kernel launcher declarations (host-side interface).
+//
+// MI355X-native hot path of the store: the reference performs one synchronous
+// MPI_Get / fi_read per row (reference: include/ddstore.hpp:229-237,
+// src/common.cxx:332-343); here a whole minibatch of rows is gathered by ONE
+// kernel launch that reads local + xGMI-peer HBM pointers directly and packs
+// (optionally dtype-casting) into a contiguous output buffer.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+// Element dtype enum shared between host dispatch and kernels.
+// bool is stored/moved as u8 (values 0/1), matching NumPy's memory layout.
+enum DDSType : int {
+    DDS_U8 = 0,
+    DDS_I32 = 1,
+    DDS_I64 = 2,
+    DDS_F32 = 3,
+    DDS_F64 = 4,
+    DDS_F16 = 5,
+    DDS_BF16 = 6,
+    DDS_NUM_TYPES = 7
+};
+
+// Max world size the in-kernel directory supports (fits in LDS; one node of
+// 8 GPUs is the design point, 128 leaves headroom for oversubscribed tests).
+#define DDS_MAX_PARTS 128
+
+namespace ddstore {
+
+// Gather `nidx` fixed-stride rows (row = `row_elems` elements of dtype
+// `in_t`) addressed by global row ids `d_idx` from the sharded store
+// described by (`d_peer_base`, `d_prefix`, `nparts`), packing them
+// contiguously into `d_out` with dtype `out_t`.
+//   d_peer_base : device array[nparts]   -- shard base pointers (self + IPC peers)
+//   d_prefix    : device array[nparts+1] -- global row prefix sums, prefix[0]=0
+void gather_rows(hipStream_t stream,
+                 const void* const* d_peer_base,
+                 const int64_t* d_prefix, int nparts,
+                 const int64_t* d_idx, int64_t nidx,
+                 int64_t row_elems, int in_t, int out_t,
+                 void* d_out);
+
+// CSR (variable-length record) gather: sample `g` owns elements
+// [d_goff[g], d_goff[g+1]) of the global element space; each element is
+// `elem_bytes` bytes (= disp * itemsize). Output element offsets per sample
+// are provided in d_out_off (exclusive scan, [nidx+1]).
+//   d_sample_prefix : array[nparts+1] -- per-rank sample-count prefix sums
+//   d_elem_prefix   : array[nparts+1] -- per-rank element-count prefix sums
+void gather_csr(hipStream_t stream,
+                const void* const* d_peer_base,
+                const int64_t* d_sample_prefix,
+                const int64_t* d_elem_prefix, int nparts,
+                const int64_t* d_goff,
+                const int64_t* d_idx, int64_t nidx,
+                const int64_t* d_out_off,
+                int64_t elem_bytes,
+                void* d_out);
+
+// Scatter rows of a packed buffer into the local shard at arbitrary local row
+// ids (inverse of gather_rows with nparts==1). Used by the epoch reshuffle to
+// place all-to-all-received rows.
+void scatter_rows_local(hipStream_t stream,
+                        void* d_base, int64_t row_elems, int elem_t,
+                        const int64_t* d_local_idx, int64_t nidx,
+                        const void* d_src);
+
+} // namespace ddstore

@@ -1,0 +1,301 @@
+// ddstore_amd CDNA4 (gfx950) gather/pack kernels.
+//
+// Design notes (MI355X):
+//  - These kernels are bandwidth-bound data movers, no MFMA. The store's
+//    reference counterpart moved one row per blocking MPI_Get/fi_read
+//    (reference: include/ddstore.hpp:229-237, src/common.cxx:332-343). Here
+//    one launch moves a whole minibatch: peer shards are directly readable
+//    through hipIpc-mapped pointers, so a remote row is a plain vector load
+//    that the fabric routes over xGMI.
+//  - 16 B/lane loads (uint4) on the fast path: 64-lane wave x 16 B = 1 KiB
+//    per instruction, the gfx950 coalescing sweet spot.
+//  - The owner directory (world_size+1 prefix sums, <=8 on one node) is
+//    staged in LDS; owner lookup is a ~3-7 step binary search over LDS.
+//  - Grids are grid-stride with a cap: >>256 workgroups to fill 8 XCDs,
+//    capped so small launches don't pay dispatch for idle blocks.
+
+#include "ddstore_kernels.h"
+
+#include <hip/hip_fp16.h>
+#include <hip/hip_bf16.h>
+#include <type_traits>
+
+namespace ddstore {
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int64_t kMaxBlocks = 32768;   // 128 blocks/CU worth of headroom
+
+inline int n_blocks(int64_t work_items) {
+    int64_t b = (work_items + kBlock - 1) / kBlock;
+    if (b < 1) b = 1;
+    if (b > kMaxBlocks) b = kMaxBlocks;
+    return (int)b;
+}
+
+__device__ __forceinline__ int owner_of(const int64_t* prefix, int nparts, int64_t row) {
+    // prefix[0] = 0; find p with prefix[p] <= row < prefix[p+1].
+    // (reference equivalent: linear `sortedsearch`, src/ddstore.cxx:5-17)
+    int lo = 0, hi = nparts - 1;
+    while (lo < hi) {
+        int mid = (lo + hi + 1) >> 1;
+        if (prefix[mid] <= row) lo = mid; else hi = mid - 1;
+    }
+    return lo;
+}
+
+template <typename Tout, typename Tin>
+__device__ __forceinline__ Tout cvt(Tin v) {
+    if constexpr (std::is_same_v<Tout, Tin>) {
+        return v;
+    } else if constexpr (std::is_same_v<Tin, __half>) {
+        return cvt<Tout, float>(__half2float(v));
+    } else if constexpr (std::is_same_v<Tin, __hip_bfloat16>) {
+        return cvt<Tout, float>(__bfloat162float(v));
+    } else if constexpr (std::is_same_v<Tout, __half>) {
+        return __float2half(static_cast<float>(v));
+    } else if constexpr (std::is_same_v<Tout, __hip_bfloat16>) {
+        return __float2bfloat16(static_cast<float>(v));
+    } else {
+        return static_cast<Tout>(v);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fixed-stride row gather, same dtype, row size a multiple of 16 B.
+// Thread t moves 16-B chunk (t % chunks_per_row) of row (t / chunks_per_row):
+// consecutive lanes walk consecutive chunks of one row -> coalesced loads and
+// stores; the per-thread i64 divide is hidden under the memory traffic.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(kBlock)
+k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int nparts,
+                  const int64_t* idx, int64_t nidx, int64_t chunks_per_row,
+                  uint4* __restrict__ out) {
+    __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
+    __shared__ const uint4* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const uint4*>(peer_base[i]);
+    __syncthreads();
+
+    const int64_t total = nidx * chunks_per_row;
+    for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
+         t += (int64_t)gridDim.x * kBlock) {
+        const int64_t r = t / chunks_per_row;
+        const int64_t c = t - r * chunks_per_row;
+        const int64_t g = idx[r];
+        const int p = owner_of(s_prefix, nparts, g);
+        out[t] = s_base[p][(g - s_prefix[p]) * chunks_per_row + c];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fixed-stride row gather, general: per-element loop with dtype cast.
+// ---------------------------------------------------------------------------
+template <typename Tin, typename Tout>
+__global__ void __launch_bounds__(kBlock)
+k_gather_rows_cast(const void* const* peer_base, const int64_t* gprefix, int nparts,
+                   const int64_t* idx, int64_t nidx, int64_t row_elems,
+                   Tout* __restrict__ out) {
+    __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
+    __shared__ const Tin* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const Tin*>(peer_base[i]);
+    __syncthreads();
+
+    const int64_t total = nidx * row_elems;
+    for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
+         t += (int64_t)gridDim.x * kBlock) {
+        const int64_t r = t / row_elems;
+        const int64_t c = t - r * row_elems;
+        const int64_t g = idx[r];
+        const int p = owner_of(s_prefix, nparts, g);
+        out[t] = cvt<Tout>(s_base[p][(g - s_prefix[p]) * row_elems + c]);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// CSR variable-length gather: one workgroup per sample (grid-stride over
+// samples); within a sample all 256 threads stream the payload with chunked
+// loads (T = uint4 / uint32_t / uint8_t picked host-side from elem_bytes
+// alignment). Samples never cross a shard boundary (enforced at add_csr).
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void __launch_bounds__(kBlock)
+k_gather_csr(const void* const* peer_base,
+             const int64_t* sample_prefix, const int64_t* elem_prefix, int nparts,
+             const int64_t* goff,
+             const int64_t* idx, int64_t nidx,
+             const int64_t* out_off, int64_t chunks_per_elem,
+             T* __restrict__ out) {
+    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
+    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
+    __shared__ const T* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
+        s_sprefix[i] = sample_prefix[i];
+        s_eprefix[i] = elem_prefix[i];
+    }
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const T*>(peer_base[i]);
+    __syncthreads();
+
+    for (int64_t s = blockIdx.x; s < nidx; s += gridDim.x) {
+        const int64_t g = idx[s];
+        const int p = owner_of(s_sprefix, nparts, g);
+        const int64_t e0 = goff[g];
+        const int64_t nch = (goff[g + 1] - e0) * chunks_per_elem;
+        const T* src = s_base[p] + (e0 - s_eprefix[p]) * chunks_per_elem;
+        T* dst = out + out_off[s] * chunks_per_elem;
+        for (int64_t c = threadIdx.x; c < nch; c += kBlock) dst[c] = src[c];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Local scatter (reshuffle placement): row r of src -> local row
+// local_idx[r] of base. Same chunk mapping as gather.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(kBlock)
+k_scatter_rows_b16(uint4* __restrict__ base, int64_t chunks_per_row,
+                   const int64_t* local_idx, int64_t nidx,
+                   const uint4* __restrict__ src) {
+    const int64_t total = nidx * chunks_per_row;
+    for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
+         t += (int64_t)gridDim.x * kBlock) {
+        const int64_t r = t / chunks_per_row;
+        const int64_t c = t - r * chunks_per_row;
+        base[local_idx[r] * chunks_per_row + c] = src[t];
+    }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(kBlock)
+k_scatter_rows_elem(T* __restrict__ base, int64_t row_elems,
+                    const int64_t* local_idx, int64_t nidx,
+                    const T* __restrict__ src) {
+    const int64_t total = nidx * row_elems;
+    for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
+         t += (int64_t)gridDim.x * kBlock) {
+        const int64_t r = t / row_elems;
+        const int64_t c = t - r * row_elems;
+        base[local_idx[r] * row_elems + c] = src[t];
+    }
+}
+
+inline int dds_itemsize(int t) {
+    switch (t) {
+        case DDS_U8: return 1;
+        case DDS_F16: case DDS_BF16: return 2;
+        case DDS_I32: case DDS_F32: return 4;
+        default: return 8;
+    }
+}
+
+template <typename Tin>
+void launch_gather_cast_out(hipStream_t stream, const void* const* pb,
+                            const int64_t* pf, int np, const int64_t* idx,
+                            int64_t n, int64_t re, int out_t, void* out, int grid) {
+    switch (out_t) {
+#define DDS_OUT(tag, T)                                                               \
+    case tag:                                                                         \
+        hipLaunchKernelGGL((k_gather_rows_cast<Tin, T>), dim3(grid), dim3(kBlock), 0, \
+                           stream, pb, pf, np, idx, n, re, (T*)out);                  \
+        break;
+        DDS_OUT(DDS_U8, uint8_t)
+        DDS_OUT(DDS_I32, int32_t)
+        DDS_OUT(DDS_I64, int64_t)
+        DDS_OUT(DDS_F32, float)
+        DDS_OUT(DDS_F64, double)
+        DDS_OUT(DDS_F16, __half)
+        DDS_OUT(DDS_BF16, __hip_bfloat16)
+#undef DDS_OUT
+    }
+}
+
+} // namespace
+
+void gather_rows(hipStream_t stream,
+                 const void* const* d_peer_base,
+                 const int64_t* d_prefix, int nparts,
+                 const int64_t* d_idx, int64_t nidx,
+                 int64_t row_elems, int in_t, int out_t,
+                 void* d_out) {
+    if (nidx == 0 || row_elems == 0) return;
+    const int64_t row_bytes = row_elems * dds_itemsize(in_t);
+    if (in_t == out_t && row_bytes % 16 == 0) {
+        const int64_t cpr = row_bytes / 16;
+        const int grid = n_blocks(nidx * cpr);
+        hipLaunchKernelGGL(k_gather_rows_b16, dim3(grid), dim3(kBlock), 0, stream,
+                           d_peer_base, d_prefix, nparts, d_idx, nidx, cpr,
+                           (uint4*)d_out);
+        return;
+    }
+    const int grid = n_blocks(nidx * row_elems);
+    switch (in_t) {
+#define DDS_IN(tag, T)                                                           \
+    case tag:                                                                    \
+        launch_gather_cast_out<T>(stream, d_peer_base, d_prefix, nparts, d_idx,  \
+                                  nidx, row_elems, out_t, d_out, grid);          \
+        break;
+        DDS_IN(DDS_U8, uint8_t)
+        DDS_IN(DDS_I32, int32_t)
+        DDS_IN(DDS_I64, int64_t)
+        DDS_IN(DDS_F32, float)
+        DDS_IN(DDS_F64, double)
+        DDS_IN(DDS_F16, __half)
+        DDS_IN(DDS_BF16, __hip_bfloat16)
+#undef DDS_IN
+    }
+}
+
+void gather_csr(hipStream_t stream,
+                const void* const* d_peer_base,
+                const int64_t* d_sample_prefix,
+                const int64_t* d_elem_prefix, int nparts,
+                const int64_t* d_goff,
+                const int64_t* d_idx, int64_t nidx,
+                const int64_t* d_out_off,
+                int64_t elem_bytes,
+                void* d_out) {
+    if (nidx == 0) return;
+    int grid = (int)(nidx < kMaxBlocks ? nidx : kMaxBlocks);
+    if (elem_bytes % 16 == 0) {
+        hipLaunchKernelGGL((k_gather_csr<uint4>), dim3(grid), dim3(kBlock), 0, stream,
+                           d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
+                           d_goff, d_idx, nidx, d_out_off, elem_bytes / 16,
+                           (uint4*)d_out);
+    } else if (elem_bytes % 4 == 0) {
+        hipLaunchKernelGGL((k_gather_csr<uint32_t>), dim3(grid), dim3(kBlock), 0, stream,
+                           d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
+                           d_goff, d_idx, nidx, d_out_off, elem_bytes / 4,
+                           (uint32_t*)d_out);
+    } else {
+        hipLaunchKernelGGL((k_gather_csr<uint8_t>), dim3(grid), dim3(kBlock), 0, stream,
+                           d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
+                           d_goff, d_idx, nidx, d_out_off, elem_bytes,
+                           (uint8_t*)d_out);
+    }
+}
+
+void scatter_rows_local(hipStream_t stream,
+                        void* d_base, int64_t row_elems, int elem_t,
+                        const int64_t* d_local_idx, int64_t nidx,
+                        const void* d_src) {
+    if (nidx == 0 || row_elems == 0) return;
+    const int64_t row_bytes = row_elems * dds_itemsize(elem_t);
+    if (row_bytes % 16 == 0) {
+        const int64_t cpr = row_bytes / 16;
+        const int grid = n_blocks(nidx * cpr);
+        hipLaunchKernelGGL(k_scatter_rows_b16, dim3(grid), dim3(kBlock), 0, stream,
+                           (uint4*)d_base, cpr, d_local_idx, nidx, (const uint4*)d_src);
+    } else {
+        const int64_t total_bytes = nidx * row_bytes;
+        const int grid = n_blocks(total_bytes);
+        hipLaunchKernelGGL((k_scatter_rows_elem<uint8_t>), dim3(grid), dim3(kBlock), 0,
+                           stream, (uint8_t*)d_base, row_bytes, d_local_idx, nidx,
+                           (const uint8_t*)d_src);
+    }
+}
+
+} // namespace ddstore

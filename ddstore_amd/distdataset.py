@@ -1,0 +1,112 @@
+"""torch Dataset adapter over a DDStore.
+
+Capability-parity rebuild of the reference adapter
+(reference: examples/vae/distdataset.py:1-93): the full dataset is split
+contiguously across the store group, each rank registers its slice, and any
+rank can read any sample through the store. Two known reference defects are
+fixed rather than replicated: the element-offset read bug
+(distdataset.py:84 fetched elements ``[idx, idx+784)`` instead of the
+sample's slice) and the missing ``import os`` -- here samples are stored as
+proper ``(n, prod(shape))`` rows so ``__getitem__(i)`` addresses row ``i``
+directly, and ``ddstore_width`` is a real constructor argument (the reference
+README documents it but the binding drops it, README.md:71-77 vs
+pyddstore.pyx:61).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Sequence
+
+import numpy as np
+import torch
+from torch.utils.data import Dataset
+
+from .comm import as_comm
+from .prefetch import PrefetchLoader
+from .store import DDStore
+
+
+def nsplit(n: int, parts: int) -> list:
+    """Contiguous split of ``n`` items into ``parts`` chunk sizes, remainder
+    spread over the leading chunks (reference distdataset.py:9-11)."""
+    base, rem = divmod(n, parts)
+    return [base + (1 if i < rem else 0) for i in range(parts)]
+
+
+class DistDataset(Dataset):
+    """Map-style Dataset over a sharded store (global index space).
+
+    Parameters mirror the reference: ``data`` and ``labels`` are the FULL
+    dataset (every rank passes the same arrays); each rank keeps only its
+    contiguous slice in the store. ``__getitem__`` works for ANY global
+    index -- remote samples are one-sided reads.
+    """
+
+    def __init__(
+        self,
+        data,
+        labels,
+        comm=None,
+        ddstore_width: Optional[int] = None,
+        device=None,
+        label: str = "train",
+        method: int = 0,
+    ):
+        data = torch.as_tensor(np.ascontiguousarray(data) if isinstance(data, np.ndarray) else data)
+        labels = torch.as_tensor(np.ascontiguousarray(labels) if isinstance(labels, np.ndarray) else labels)
+        if labels.dim() == 1:
+            labels = labels.unsqueeze(1)
+        assert data.shape[0] == labels.shape[0], "data/labels length mismatch"
+        self.total = int(data.shape[0])
+        self.sample_shape = tuple(data.shape[1:])
+        self.ddstore = DDStore(comm, method=method, device=device, ddstore_width=ddstore_width)
+        rank, size = self.ddstore.rank, self.ddstore.size
+        counts = nsplit(self.total, size)
+        lo = sum(counts[:rank])
+        hi = lo + counts[rank]
+        self.label = label
+        self._data_var = f"{label}data"
+        self._label_var = f"{label}labels"
+        flat = data[lo:hi].reshape(hi - lo, -1).contiguous()
+        self.ddstore.add(self._data_var, flat)
+        self.ddstore.add(self._label_var, labels[lo:hi].contiguous())
+        self.data_dtype = flat.dtype
+        self.label_dtype = labels.dtype
+
+    def __len__(self) -> int:
+        return self.total
+
+    def __getitem__(self, idx: int):
+        x = self.ddstore.get_batch(self._data_var, [int(idx)])
+        y = self.ddstore.get_batch(self._label_var, [int(idx)])
+        return x.view(*self.sample_shape), y.view(-1)[0]
+
+    def loader(
+        self,
+        indices: Sequence[int],
+        batch_size: int,
+        out_dtype: Optional[torch.dtype] = None,
+        depth: int = 2,
+        drop_last: bool = False,
+    ) -> PrefetchLoader:
+        """Fast path: side-stream prefetched minibatches (data, labels) for a
+        per-epoch index order (e.g. from a DistributedSampler)."""
+        return PrefetchLoader(
+            self.ddstore,
+            self._data_var,
+            indices,
+            batch_size,
+            out_dtype=out_dtype,
+            label_name=self._label_var,
+            depth=depth,
+            drop_last=drop_last,
+        )
+
+    def epoch_begin(self):
+        self.ddstore.epoch_begin()
+
+    def epoch_end(self):
+        self.ddstore.epoch_end()
+
+    def free(self):
+        self.ddstore.free()

@@ -1,0 +1,133 @@
+"""Side-stream minibatch prefetcher.
+
+The reference has no prefetch: its DataLoader path issues one blocking
+MPI_Get per sample (reference call stack SURVEY §3.2/§3.3). Here the next
+minibatch is gathered by the CDNA4 gather kernel on a dedicated side HIP
+stream, double-buffered, so the training step on the main stream never waits
+for sample fetch (BASELINE config 5). Synchronization is via HIP events:
+
+    side stream:  [wait buf free] -> gather batch k -> record ready_k
+    main stream:  wait ready_k -> train step on batch k -> record free_k
+"""
+from __future__ import annotations
+
+from typing import Iterator, Optional, Sequence, Union
+
+import numpy as np
+import torch
+
+from .store import DDStore
+
+
+class PrefetchLoader:
+    """Iterate minibatches of ``store[name]`` (and optionally a second
+    aligned variable, e.g. labels) for a given epoch index order.
+
+    Parameters
+    ----------
+    store : DDStore
+    name : str            main (data) variable
+    indices : 1-D int sequence/tensor of global row ids for this epoch
+              (e.g. the ids a DistributedSampler assigned to this rank)
+    batch_size : int      rows per yielded batch
+    out_dtype : optional  fused dtype cast in the gather kernel
+    label_name : optional second variable gathered with the same indices
+    depth : int           ring depth (2 = classic double buffering)
+    drop_last : bool
+    """
+
+    def __init__(
+        self,
+        store: DDStore,
+        name: str,
+        indices: Union[Sequence[int], torch.Tensor, np.ndarray],
+        batch_size: int,
+        out_dtype: Optional[torch.dtype] = None,
+        label_name: Optional[str] = None,
+        label_dtype: Optional[torch.dtype] = None,
+        depth: int = 2,
+        drop_last: bool = False,
+    ):
+        self.store = store
+        self.name = name
+        self.label_name = label_name
+        self.out_dtype = out_dtype
+        self.label_dtype = label_dtype
+        self.batch_size = int(batch_size)
+        self.depth = max(2, int(depth))
+        self.drop_last = drop_last
+        idx = torch.as_tensor(indices, dtype=torch.int64).flatten()
+        self.indices = idx
+
+    def __len__(self) -> int:
+        n = self.indices.numel()
+        return n // self.batch_size if self.drop_last else (n + self.batch_size - 1) // self.batch_size
+
+    def _batches(self, idx: torch.Tensor):
+        n = idx.numel()
+        stop = (n // self.batch_size) * self.batch_size if self.drop_last else n
+        return [idx[i : min(i + self.batch_size, stop)] for i in range(0, stop, self.batch_size)]
+
+    def __iter__(self) -> Iterator:
+        if self.store.mode != "hip":
+            for b in self._batches(self.indices):
+                data = self.store.get_batch(self.name, b, dtype=self.out_dtype)
+                if self.label_name is None:
+                    yield data
+                else:
+                    yield data, self.store.get_batch(self.label_name, b, dtype=self.label_dtype)
+            return
+        yield from self._iter_hip()
+
+    def _iter_hip(self) -> Iterator:
+        store = self.store
+        device = store.device
+        side = torch.cuda.Stream(device)
+        idx_dev = self.indices.to(device, non_blocking=False)
+        batches = self._batches(idx_dev)
+        nb = len(batches)
+        slots = [
+            {
+                "data": None,
+                "label": None,
+                "ready": torch.cuda.Event(),
+                "free": torch.cuda.Event(),
+                "free_recorded": False,
+            }
+            for _ in range(self.depth)
+        ]
+
+        def launch(j: int):
+            slot = slots[j % self.depth]
+            with torch.cuda.stream(side):
+                if slot["free_recorded"]:
+                    side.wait_event(slot["free"])
+                slot["data"] = store.get_batch(
+                    self.name, batches[j], out=slot["data"], dtype=self.out_dtype
+                )
+                if self.label_name is not None:
+                    slot["label"] = store.get_batch(
+                        self.label_name, batches[j], out=slot["label"],
+                        dtype=self.label_dtype,
+                    )
+                slot["ready"].record(side)
+
+        for j in range(min(self.depth, nb)):
+            launch(j)
+        cur = torch.cuda.current_stream(device)
+        for i in range(nb):
+            slot = slots[i % self.depth]
+            cur = torch.cuda.current_stream(device)
+            cur.wait_event(slot["ready"])
+            if self.label_name is None:
+                yield slot["data"]
+            else:
+                yield slot["data"], slot["label"]
+            # the consumer has enqueued its use of the buffers on the current
+            # stream by the time it asks for the next batch
+            cur = torch.cuda.current_stream(device)
+            slot["free"].record(cur)
+            slot["free_recorded"] = True
+            nxt = i + self.depth
+            if nxt < nb:
+                launch(nxt)

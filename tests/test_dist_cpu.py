@@ -1,0 +1,239 @@
+"""Multi-process (gloo, CPU) store tests -- the distributed semantics the
+reference exercises under ``mpirun`` (test/demo.py, test/test.py),
+run here with spawned torch.distributed ranks on 127.0.0.1.
+
+Self-verifying data pattern: rank r fills its shard with the constant r+1 so
+a fetched row proves who owned it (reference test/demo.py:35-39)."""
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+
+from tests.dist_utils import run_dist
+
+pytestmark = pytest.mark.timeout(300)
+
+NUM, DIM = 256, 16
+
+
+def _mkstore(rank, world, width=None):
+    from ddstore_amd import DDStore
+
+    return DDStore(device="cpu", ddstore_width=width)
+
+
+# --------------------------------------------------------------------------
+def _w_demo(rank, world):
+    """demo.py parity: add -> epoched random gets -> value check -> free."""
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    arr = np.full((NUM, DIM), rank + 1, dtype=np.float64)
+    s.add("x", arr)
+    rng = np.random.default_rng(1000 + rank)
+    for _ in range(8):
+        s.epoch_begin()
+        idx = int(rng.integers(0, NUM * world))
+        out = np.zeros((1, DIM), dtype=np.float64)
+        s.get("x", out, start=idx)
+        assert out.mean() == idx // NUM + 1, (idx, out.mean())
+        s.epoch_end()
+    s.free()
+
+
+def test_demo_parity_ws2():
+    run_dist(_w_demo, 2)
+
+
+def test_demo_parity_ws4():
+    run_dist(_w_demo, 4)
+
+
+# --------------------------------------------------------------------------
+def _w_batch(rank, world):
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    arr = np.full((NUM, DIM), rank + 1, dtype=np.float32)
+    s.add("x", arr)
+    rng = np.random.default_rng(7)  # same on all ranks
+    idx = rng.integers(0, NUM * world, size=64)
+    out = s.get_batch("x", idx)
+    expect = (idx // NUM + 1).astype(np.float32)
+    assert np.array_equal(out.numpy()[:, 0], expect)
+    # global shuffle sum invariant
+    total = out.numpy().sum()
+    expected = expect.sum() * DIM
+    assert total == expected
+    s.free()
+
+
+def test_get_batch_global(ws=2):
+    run_dist(_w_batch, ws)
+
+
+# --------------------------------------------------------------------------
+def _w_boundary(rank, world):
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    s.add("x", np.zeros((NUM, DIM), dtype=np.float32))
+    out = np.zeros((2, DIM), dtype=np.float32)
+    # crossing the shard boundary must throw (reference ddstore.hpp:210-214)
+    try:
+        s.get("x", out, start=NUM - 1)
+        raise AssertionError("expected Invalid count on target")
+    except RuntimeError as e:
+        assert "Invalid count on target" in str(e)
+    s.free()
+
+
+def test_cross_shard_get_raises():
+    run_dist(_w_boundary, 2)
+
+
+# --------------------------------------------------------------------------
+def _w_csr(rank, world):
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    # rank r owns samples with lengths [r+1, r+2], values == global sample id
+    lengths = [rank + 1, rank + 2]
+    nelem = sum(lengths)
+    gid0 = 2 * rank
+    vals = np.concatenate(
+        [np.full(l, gid0 + i, dtype=np.float32) for i, l in enumerate(lengths)]
+    ).reshape(-1, 1)
+    s.add_csr("c", vals, lengths)
+    # every rank reads every sample
+    ntotal = 2 * world
+    v, off = s.get_csr("c", list(range(ntotal)))
+    off = off.tolist()
+    for g in range(ntotal):
+        seg = v[off[g] : off[g + 1], 0].numpy()
+        expected_len = (g // 2) + 1 + (g % 2)
+        assert len(seg) == expected_len, (g, len(seg), expected_len)
+        assert (seg == g).all()
+    s.free()
+
+
+def test_csr_remote():
+    run_dist(_w_csr, 3)
+
+
+# --------------------------------------------------------------------------
+def _w_reshuffle(rank, world):
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    base = np.arange(rank * NUM, (rank + 1) * NUM, dtype=np.float32)
+    arr = np.repeat(base[:, None], DIM, axis=1)  # row g == value g
+    s.add("x", arr)
+    s.reshuffle("x", seed=99)
+    perm = np.random.default_rng(99).permutation(NUM * world)
+    out = s.get_batch("x", list(range(NUM * world)))
+    assert np.array_equal(out.numpy()[:, 0], perm.astype(np.float32))
+    s.free()
+
+
+def test_reshuffle_ws2():
+    run_dist(_w_reshuffle, 2)
+
+
+def test_reshuffle_ws4():
+    run_dist(_w_reshuffle, 4)
+
+
+# --------------------------------------------------------------------------
+def _w_coexist(rank, world):
+    """Store traffic interleaved with a DDP-style all_reduce per batch
+    (reference test/test.py:153-154 coexistence probe)."""
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    arr = np.full((NUM, DIM), rank + 1, dtype=np.float32)
+    s.add("a", arr)
+    s.add("b", 10 * arr)
+    rng = np.random.default_rng(5)
+    for _ in range(4):
+        s.epoch_begin()
+        idx = rng.integers(0, NUM * world, size=8)
+        oa = s.get_batch("a", idx)
+        ob = s.get_batch("b", idx)
+        assert np.array_equal(ob.numpy(), 10 * oa.numpy())
+        s.epoch_end()
+        t = torch.tensor([float(rank)])
+        dist.all_reduce(t)
+        assert t.item() == world * (world - 1) / 2
+    s.free()
+
+
+def test_store_ddp_coexistence():
+    run_dist(_w_coexist, 2)
+
+
+# --------------------------------------------------------------------------
+def _w_width(rank, world):
+    """Replication groups: width=2 on 4 ranks -> two groups, each holding a
+    full replica partitioned internally (reference README.md:154-172)."""
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu", ddstore_width=2)
+    assert s.size == 2 and s.rank == rank % 2
+    arr = np.full((NUM, DIM), s.rank + 1, dtype=np.float32)
+    s.add("x", arr)
+    q = s.query("x")
+    assert q["nrows_total"] == 2 * NUM
+    out = np.zeros((1, DIM), dtype=np.float32)
+    s.get("x", out, start=NUM + 3)  # owned by group-rank 1
+    assert (out == 2).all()
+    s.free()
+
+
+def test_ddstore_width_groups():
+    run_dist(_w_width, 4)
+
+
+# --------------------------------------------------------------------------
+def _w_uniform_disp(rank, world):
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    dim = 4 if rank == 0 else 8
+    try:
+        s.add("x", np.zeros((4, dim), dtype=np.float32))
+        raise AssertionError("expected disp validation error")
+    except ValueError as e:
+        assert "uniform" in str(e)
+
+
+def test_disp_uniform_validation():
+    run_dist(_w_uniform_disp, 2)
+
+
+# --------------------------------------------------------------------------
+def _w_distdataset(rank, world):
+    from ddstore_amd import DistDataset
+
+    n = 40
+    data = np.arange(n * 6, dtype=np.float32).reshape(n, 2, 3)
+    labels = np.arange(n, dtype=np.int64)
+    ds = DistDataset(data, labels, device="cpu")
+    assert len(ds) == n
+    for g in [0, n - 1, n // 2 + 1]:
+        x, y = ds[g]
+        assert x.shape == (2, 3)
+        assert np.array_equal(x.numpy(), data[g])
+        assert y.item() == g
+    # prefetch-loader path with a shuffled global order
+    order = np.random.default_rng(3).permutation(n)
+    got = []
+    for xb, yb in ds.loader(order, batch_size=7):
+        assert xb.shape[1] == 6
+        got.extend(yb.view(-1).tolist())
+    assert got == order.tolist()
+    ds.free()
+
+
+def test_distdataset_ws2():
+    run_dist(_w_distdataset, 2)

@@ -1,0 +1,169 @@
+"""Single-rank CPU store tests: API surface, dtype dispatch, error semantics.
+
+Mirrors the reference's self-verifying synthetic-data philosophy
+plus the error paths the reference defines but never tests (ddstore.hpp:81-82,
+:202-203, :210-214; ddstore.cxx:57, :71).
+"""
+import numpy as np
+import pytest
+import torch
+
+from ddstore_amd import DDStore
+
+# the six reference dtypes (pyddstore.pyx:69-82) plus the new half types
+REF_DTYPES = [np.int32, np.int64, np.uint8, np.float32, np.float64, np.bool_]
+
+
+@pytest.fixture
+def store():
+    s = DDStore(device="cpu")
+    yield s
+    s.free()
+
+
+@pytest.mark.parametrize("dtype", REF_DTYPES)
+def test_roundtrip_dtypes(store, dtype):
+    if dtype == np.bool_:
+        arr = (np.arange(40) % 3 == 0).reshape(10, 4)
+    else:
+        arr = np.arange(40, dtype=dtype).reshape(10, 4)
+    store.add(f"v{np.dtype(dtype).name}", arr)
+    out = np.zeros((4, 4), dtype=dtype)
+    store.get(f"v{np.dtype(dtype).name}", out, start=3)
+    assert np.array_equal(out, arr[3:7])
+
+
+@pytest.mark.parametrize("tdtype", [torch.float16, torch.bfloat16])
+def test_half_dtypes(store, tdtype):
+    arr = torch.arange(24, dtype=tdtype).reshape(6, 4)
+    store.add("h", arr)
+    out = store.get_batch("h", [5, 0, 2])
+    assert out.dtype == tdtype
+    assert torch.equal(out, arr[[5, 0, 2]])
+
+
+def test_get_batch_and_cast(store):
+    arr = np.random.rand(32, 8).astype(np.float32)
+    store.add("x", arr)
+    idx = [31, 0, 7, 7, 16]
+    out = store.get_batch("x", idx)
+    assert np.array_equal(out.numpy(), arr[idx])
+    out64 = store.get_batch("x", idx, dtype=torch.float64)
+    assert out64.dtype == torch.float64
+    assert np.allclose(out64.numpy(), arr[idx].astype(np.float64))
+
+
+def test_1d_array_disp1(store):
+    # reference: 1-D arrays get disp=1, element-addressed (pyddstore.pyx:67-68)
+    arr = np.arange(100, dtype=np.float64)
+    store.add("flat", arr)
+    q = store.query("flat")
+    assert q["disp"] == 1 and q["nrows_total"] == 100
+    out = np.zeros(7, dtype=np.float64)
+    store.get("flat", out, start=42)
+    assert np.array_equal(out, arr[42:49])
+
+
+def test_init_update(store):
+    store.init("y", 10, 4, itemsize=8)
+    out = np.zeros((10, 4), dtype=np.float64)
+    store.get("y", out, 0)
+    assert (out == 0).all()
+    store.update("y", np.full((3, 4), 5.0), offset=7)
+    store.get("y", out, 0)
+    assert (out[7:] == 5).all() and (out[:7] == 0).all()
+    # itemsize (not dtype) is what update checks -- reference ddstore.hpp:189-190
+    store.update("y", np.arange(4, dtype=np.int64).reshape(1, 4), offset=0)
+    with pytest.raises(RuntimeError, match="itemsize"):
+        store.update("y", np.zeros((1, 4), dtype=np.float32))
+
+
+def test_update_out_of_range(store):
+    store.init("z", 4, 2, itemsize=4)
+    with pytest.raises(RuntimeError, match="out of range"):
+        store.update("z", np.zeros((2, 2), dtype=np.float32), offset=3)
+
+
+def test_invalid_start(store):
+    store.add("x", np.zeros((5, 2), dtype=np.float32))
+    out = np.zeros((1, 2), dtype=np.float32)
+    with pytest.raises(RuntimeError, match="Invalid start on target"):
+        store.get("x", out, start=5)
+    with pytest.raises(RuntimeError, match="Invalid start on target"):
+        store.get("x", out, start=-1)
+    out5 = np.zeros((3, 2), dtype=np.float32)
+    with pytest.raises(RuntimeError, match="Invalid count on target"):
+        store.get("x", out5, start=3)
+
+
+def test_epoch_fsm(store):
+    # reference throws on double-begin / end-without-begin (ddstore.cxx:57,:71)
+    store.epoch_begin()
+    with pytest.raises(RuntimeError, match="epoch already began"):
+        store.epoch_begin()
+    store.epoch_end()
+    with pytest.raises(RuntimeError, match="epoch has not begun"):
+        store.epoch_end()
+
+
+def test_unknown_and_duplicate(store):
+    with pytest.raises((RuntimeError, KeyError)):
+        store.query("nope")
+    store.add("dup", np.zeros((2, 2), dtype=np.float32))
+    with pytest.raises(RuntimeError, match="already exists"):
+        store.add("dup", np.zeros((2, 2), dtype=np.float32))
+
+
+def test_get_dtype_itemsize_check(store):
+    store.add("x", np.zeros((5, 2), dtype=np.float32))
+    bad = np.zeros((2, 2), dtype=np.float64)
+    with pytest.raises(RuntimeError, match="itemsize"):
+        store.get("x", bad, 0)
+
+
+def test_csr_roundtrip(store):
+    lengths = [3, 0, 5, 2]
+    vals = np.arange(10 * 2, dtype=np.float32).reshape(10, 2)
+    store.add_csr("c", vals, lengths)
+    v, off = store.get_csr("c", [2, 0, 1, 3])
+    assert off.tolist() == [0, 5, 8, 8, 10]
+    assert np.array_equal(v[0:5].numpy(), vals[3:8])
+    assert np.array_equal(v[5:8].numpy(), vals[0:3])
+    assert np.array_equal(v[8:10].numpy(), vals[8:10])
+
+
+def test_local_shard_view(store):
+    arr = np.arange(12, dtype=np.float32).reshape(6, 2)
+    store.add("x", arr)
+    sh = store.local_shard("x")
+    assert sh.shape == (6, 2)
+    assert np.array_equal(sh.numpy(), arr)
+
+
+def test_reshuffle_single_rank(store):
+    arr = np.arange(64, dtype=np.float32).reshape(16, 4)
+    store.add("x", arr)
+    store.reshuffle("x", seed=123)
+    perm = np.random.default_rng(123).permutation(16)
+    out = store.get_batch("x", list(range(16)))
+    assert np.array_equal(out.numpy(), arr[perm])
+
+
+def test_stats(store):
+    store.add("x", np.zeros((8, 4), dtype=np.float32))
+    store.get_batch("x", [0, 1, 2])
+    st = store.stats()["x"]
+    assert st["n_gather"] == 1 and st["rows_gathered"] == 3
+    assert st["bytes_gathered"] == 3 * 4 * 4
+
+
+def test_method_param_compat():
+    # both reference transports map onto the single native path
+    s0 = DDStore(device="cpu", method=0)
+    s1 = DDStore(device="cpu", method=1)
+    for s in (s0, s1):
+        s.add("x", np.ones((4, 2), dtype=np.float32))
+        out = np.zeros((2, 2), dtype=np.float32)
+        s.get("x", out, 1)
+        assert (out == 1).all()
+        s.free()
