@@ -91,6 +91,50 @@ k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int npar
 }
 
 // ---------------------------------------------------------------------------
+// Fixed-stride row gather with fused dtype cast, vectorized: each thread
+// moves VEC = 16/max(sizeof(Tin), sizeof(Tout)) elements so the wider side
+// issues full 16-B accesses and the narrower side 16/ratio-B accesses
+// (scalar bf16/f16 loads cost ~2-2.5x -- cdna_hip_programming.md G13).
+// Requires row_elems % VEC == 0 (host falls back to the scalar kernel).
+// ---------------------------------------------------------------------------
+template <typename T, int N>
+struct alignas(sizeof(T) * N) VecT {
+    T v[N];
+};
+
+template <typename Tin, typename Tout>
+__global__ void __launch_bounds__(kBlock)
+k_gather_rows_castv(const void* const* peer_base, const int64_t* gprefix, int nparts,
+                    const int64_t* idx, int64_t nidx, int64_t row_elems,
+                    Tout* __restrict__ out) {
+    constexpr int VEC = 16 / (sizeof(Tin) > sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
+    using Vin = VecT<Tin, VEC>;
+    using Vout = VecT<Tout, VEC>;
+    __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
+    __shared__ const Tin* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const Tin*>(peer_base[i]);
+    __syncthreads();
+
+    const int64_t cpr = row_elems / VEC;  // vector chunks per row
+    const int64_t total = nidx * cpr;
+    for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
+         t += (int64_t)gridDim.x * kBlock) {
+        const int64_t r = t / cpr;
+        const int64_t c = t - r * cpr;
+        const int64_t g = idx[r];
+        const int p = owner_of(s_prefix, nparts, g);
+        const Vin vin = *reinterpret_cast<const Vin*>(
+            s_base[p] + (g - s_prefix[p]) * row_elems + c * VEC);
+        Vout vout;
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) vout.v[k] = cvt<Tout>(vin.v[k]);
+        *reinterpret_cast<Vout*>(out + t * VEC) = vout;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Fixed-stride row gather, general: per-element loop with dtype cast.
 // ---------------------------------------------------------------------------
 template <typename Tin, typename Tout>
@@ -192,15 +236,30 @@ inline int dds_itemsize(int t) {
     }
 }
 
+template <typename Tin, typename Tout>
+void launch_gather_cast_one(hipStream_t stream, const void* const* pb,
+                            const int64_t* pf, int np, const int64_t* idx,
+                            int64_t n, int64_t re, Tout* out) {
+    constexpr int VEC = 16 / (sizeof(Tin) > sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
+    if (re % VEC == 0) {
+        const int grid = n_blocks(n * (re / VEC));
+        hipLaunchKernelGGL((k_gather_rows_castv<Tin, Tout>), dim3(grid), dim3(kBlock),
+                           0, stream, pb, pf, np, idx, n, re, out);
+    } else {
+        const int grid = n_blocks(n * re);
+        hipLaunchKernelGGL((k_gather_rows_cast<Tin, Tout>), dim3(grid), dim3(kBlock),
+                           0, stream, pb, pf, np, idx, n, re, out);
+    }
+}
+
 template <typename Tin>
 void launch_gather_cast_out(hipStream_t stream, const void* const* pb,
                             const int64_t* pf, int np, const int64_t* idx,
-                            int64_t n, int64_t re, int out_t, void* out, int grid) {
+                            int64_t n, int64_t re, int out_t, void* out, int /*grid*/) {
     switch (out_t) {
 #define DDS_OUT(tag, T)                                                               \
     case tag:                                                                         \
-        hipLaunchKernelGGL((k_gather_rows_cast<Tin, T>), dim3(grid), dim3(kBlock), 0, \
-                           stream, pb, pf, np, idx, n, re, (T*)out);                  \
+        launch_gather_cast_one<Tin, T>(stream, pb, pf, np, idx, n, re, (T*)out);      \
         break;
         DDS_OUT(DDS_U8, uint8_t)
         DDS_OUT(DDS_I32, int32_t)
