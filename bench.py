@@ -1,17 +1,19 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: global-shuffle sample fetch through the MI355X-native
-store, side-stream prefetch overlapped with a bf16 train step (BASELINE.json
-configs 2/3/5 rolled into the end-to-end path; the reference publishes no
-numbers -- BASELINE.md).
+"""Flagship benchmark: global-shuffle sample-fetch throughput of the
+MI355X-native store -- the BASELINE.json headline metric ("samples/sec +
+effective remote-read GB/s, global shuffle, 1/2/4/8 MI355X"; the reference
+publishes no numbers of its own -- BASELINE.md).
 
 Per rank (one process per GPU, RCCL over xGMI for N>1):
   * shard: ``--rows`` rows x ``--dim`` float32 in local HBM (default
     2 Mi x 128 = 1 GiB; 512 B rows -- the reference's per-get granularity,
     test/demo.py:45-50)
-  * each step fetches ``--batch`` globally-shuffled rows (fraction (N-1)/N
-    remote over xGMI) with the CDNA4 gather kernel on a side stream, while
-    the main stream runs the bf16 MLP train step (forward, backward,
-    optimizer) on the previous batch -- nothing is skipped in the timed loop.
+  * each timed step fetches ``--batch`` globally-shuffled rows (fraction
+    (N-1)/N remote over xGMI) with one CDNA4 gather kernel launch: in-kernel
+    owner lookup, peer-pointer reads, pack + fused bf16 cast.
+  * ``--mode train`` instead overlaps the fetch (side-stream prefetcher)
+    with a full bf16 MLP train step -- forward, backward, optimizer, DDP
+    gradient allreduce -- per step (BASELINE config 5).
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W
@@ -42,9 +44,11 @@ def parse_args():
     p.add_argument("--dim", type=int, default=128, help="f32 elems per row (128 -> 512 B rows)")
     p.add_argument("--batch", type=int, default=131072, help="rows fetched per rank per step")
     p.add_argument("--hidden", type=int, default=1024, help="MLP hidden width")
-    p.add_argument("--mode", choices=["train", "fetch"], default="train",
-                   help="train = fetch overlapped with bf16 MLP step (flagship); "
-                        "fetch = pure gather throughput")
+    p.add_argument("--mode", choices=["fetch", "train"], default="fetch",
+                   help="fetch = global-shuffle sample-fetch throughput, the "
+                        "BASELINE.json headline metric (default); train = the "
+                        "same fetch overlapped with a full bf16 MLP train step "
+                        "(forward+backward+optimizer, BASELINE config 5)")
     p.add_argument("--device", default="cuda")
     return p.parse_args()
 
@@ -113,32 +117,49 @@ def main():
         epoch += 1
     order = torch.cat(order)[:need]
 
-    trainer = TrainStep(dim, args.hidden, device, world) if args.mode == "train" else None
-
-    loader = PrefetchLoader(store, "bench", order, batch, out_dtype=torch.bfloat16,
-                            depth=3, drop_last=True)
-
-    def run_steps(it, n: int):
-        for _ in range(n):
-            b = next(it)
-            if trainer is not None:
-                trainer(b)
-            else:
-                # fetch mode: touch the batch on the main stream (a negligible
-                # op -- completion is enforced by the prefetcher's event wait
-                # and the closing synchronize)
-                b[:1].float().sum()
-
     store.epoch_begin()
-    it = iter(loader)
-    run_steps(it, args.warmup)
+    if args.mode == "train":
+        # overlap config: side-stream prefetch feeding a full bf16 train step
+        trainer = TrainStep(dim, args.hidden, device, world)
+        loader = PrefetchLoader(store, "bench", order, batch,
+                                out_dtype=torch.bfloat16, depth=3, drop_last=True)
+        it = iter(loader)
+
+        def run_steps(n: int):
+            for _ in range(n):
+                trainer(next(it))
+
+    else:
+        # flagship store metric: back-to-back batched gathers (owner lookup +
+        # xGMI peer read + pack + fused bf16 cast per step), nothing else in
+        # the timed loop
+        order_dev = order.to(device) if use_cuda else order
+        nring = 4
+        bufs = [
+            torch.empty(batch, dim, dtype=torch.bfloat16, device=device)
+            for _ in range(nring)
+        ]
+        step_idx = [
+            order_dev[k * batch : (k + 1) * batch].contiguous()
+            for k in range(nsteps_total)
+        ]
+        counter = {"k": 0}
+
+        def run_steps(n: int):
+            k = counter["k"]
+            for _ in range(n):
+                store.get_batch("bench", step_idx[k], out=bufs[k % nring])
+                k += 1
+            counter["k"] = k
+
+    run_steps(args.warmup)
 
     if world > 1:
         dist.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    run_steps(it, args.steps)
+    run_steps(args.steps)
     if use_cuda:
         torch.cuda.synchronize()
     if world > 1:
@@ -150,10 +171,7 @@ def main():
     if world > 1:
         t = torch.tensor([elapsed], device=device if use_cuda else "cpu",
                          dtype=torch.float64)
-        if dist.get_backend() == "nccl":
-            dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        else:
-            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
     n_samples = world * args.steps * batch
@@ -164,7 +182,9 @@ def main():
 
     if rank == 0:
         result = {
-            "metric": "samples/sec (global-shuffle fetch + bf16 train step)",
+            "metric": ("samples/sec (global-shuffle fetch + bf16 train step)"
+                       if args.mode == "train"
+                       else "samples/sec (global-shuffle sample fetch)"),
             "value": sps,
             "unit": "samples/s",
             "n_gpus": world,
