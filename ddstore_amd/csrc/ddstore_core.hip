@@ -41,7 +41,16 @@
 
 #include "ddstore_kernels.h"
 
+// roctx ranges so rocprofv3 --marker-trace attributes store phases
+// (the reference has no tracing at all)
+#include <roctracer/roctx.h>
+
 namespace {
+
+struct RoctxRange {
+    explicit RoctxRange(const char* name) { roctxRangePushA(name); }
+    ~RoctxRange() { roctxRangePop(); }
+};
 
 #define HIP_CHECK(expr)                                                        \
     do {                                                                       \
@@ -122,6 +131,7 @@ struct DeviceVar {
     std::vector<char> opened;    // 1 where hipIpcOpenMemHandle was used
     void** d_peers = nullptr;
     int64_t* d_prefix = nullptr;
+    unsigned long long* d_oob = nullptr;  // in-kernel OOB-index counter
     // CSR extras
     std::vector<int64_t> elem_prefix;
     int64_t* d_elem_prefix = nullptr;
@@ -242,20 +252,24 @@ public:
                 v.opened[r] = 1;
             }
         }
-        // one device block: [peer ptrs | prefix | elem_prefix?]
+        // one device block: [peer ptrs | prefix | elem_prefix? | oob counter]
         size_t nb = nparts_ * sizeof(void*) + (nparts_ + 1) * 8 +
-                    (v.is_csr ? (nparts_ + 1) * 8 : 0);
+                    (v.is_csr ? (nparts_ + 1) * 8 : 0) + 8;
         HIP_CHECK(hipMalloc((void**)&v.d_peers, nb));
         v.d_prefix = reinterpret_cast<int64_t*>(v.d_peers + nparts_);
         HIP_CHECK(hipMemcpy(v.d_peers, v.peers.data(), nparts_ * sizeof(void*),
                             hipMemcpyHostToDevice));
         HIP_CHECK(hipMemcpy(v.d_prefix, v.prefix.data(), (nparts_ + 1) * 8,
                             hipMemcpyHostToDevice));
+        int64_t* tail = v.d_prefix + nparts_ + 1;
         if (v.is_csr) {
-            v.d_elem_prefix = v.d_prefix + nparts_ + 1;
+            v.d_elem_prefix = tail;
             HIP_CHECK(hipMemcpy(v.d_elem_prefix, v.elem_prefix.data(),
                                 (nparts_ + 1) * 8, hipMemcpyHostToDevice));
+            tail += nparts_ + 1;
         }
+        v.d_oob = reinterpret_cast<unsigned long long*>(tail);
+        HIP_CHECK(hipMemset(v.d_oob, 0, 8));
     }
 
     void update(const std::string& name, const at::Tensor& src, int64_t offset) {
@@ -306,6 +320,7 @@ public:
 
     void gather(const std::string& name, const at::Tensor& idx, at::Tensor out) {
         // the batched hot path: one launch per minibatch
+        RoctxRange rr_("ddstore::gather");
         DeviceVar& v = var(name);
         TORCH_CHECK(!v.is_csr, "ddstore gather: use gather_csr for CSR variables");
         check_peers(v);
@@ -319,7 +334,7 @@ public:
         if (dds_itemsize(out_t) == v.itemsize && out_t != v.dds_t) out_t = v.dds_t;
         ddstore::gather_rows(stream(), (const void* const*)v.d_peers, v.d_prefix,
                              nparts_, idx.data_ptr<int64_t>(), nidx, v.row_elems,
-                             v.dds_t, out_t, out.data_ptr());
+                             v.dds_t, out_t, out.data_ptr(), v.d_oob);
         v.n_gather += 1;
         v.rows_gathered += nidx;
         v.bytes_gathered += nidx * v.row_elems * v.itemsize;
@@ -327,6 +342,7 @@ public:
 
     void gather_csr(const std::string& name, const at::Tensor& idx,
                     const at::Tensor& out_off, at::Tensor out, int64_t total_elems) {
+        RoctxRange rr_("ddstore::gather_csr");
         DeviceVar& v = var(name);
         TORCH_CHECK(v.is_csr, "ddstore gather_csr: not a CSR variable");
         check_peers(v);
@@ -343,7 +359,7 @@ public:
                             v.d_elem_prefix, nparts_, v.d_goff,
                             idx.data_ptr<int64_t>(), nidx,
                             out_off.data_ptr<int64_t>(),
-                            v.row_elems * v.itemsize, out.data_ptr());
+                            v.row_elems * v.itemsize, out.data_ptr(), v.d_oob);
         v.n_gather += 1;
         v.rows_gathered += nidx;
         v.bytes_gathered += total_elems * v.row_elems * v.itemsize;
@@ -360,9 +376,10 @@ public:
                     "ddstore scatter_local: bad src");
         TORCH_CHECK(src.numel() == local_idx.numel() * v.row_elems,
                     "ddstore scatter_local: shape mismatch");
-        ddstore::scatter_rows_local(stream(), v.base, v.row_elems, v.dds_t,
-                                    local_idx.data_ptr<int64_t>(),
-                                    local_idx.numel(), src.data_ptr());
+        check_peers(v);  // d_oob lives in the metadata block
+        ddstore::scatter_rows_local(stream(), v.base, v.nrows_local, v.row_elems,
+                                    v.dds_t, local_idx.data_ptr<int64_t>(),
+                                    local_idx.numel(), src.data_ptr(), v.d_oob);
     }
 
     // Zero-copy view of the local shard as a torch tensor (does NOT own).
@@ -398,6 +415,12 @@ public:
         d["n_gather"] = v.n_gather;
         d["rows_gathered"] = v.rows_gathered;
         d["bytes_gathered"] = v.bytes_gathered;
+        if (v.d_oob) {
+            unsigned long long oob = 0;
+            HIP_CHECK(hipStreamSynchronize(stream()));
+            HIP_CHECK(hipMemcpy(&oob, v.d_oob, 8, hipMemcpyDeviceToHost));
+            d["oob_skipped"] = (int64_t)oob;
+        }
         return d;
     }
 

@@ -35,12 +35,14 @@ namespace ddstore {
 // contiguously into `d_out` with dtype `out_t`.
 //   d_peer_base : device array[nparts]   -- shard base pointers (self + IPC peers)
 //   d_prefix    : device array[nparts+1] -- global row prefix sums, prefix[0]=0
+// d_oob: device counter incremented once per out-of-range index (the row is
+// skipped instead of read out of bounds; checked lazily via query()).
 void gather_rows(hipStream_t stream,
                  const void* const* d_peer_base,
                  const int64_t* d_prefix, int nparts,
                  const int64_t* d_idx, int64_t nidx,
                  int64_t row_elems, int in_t, int out_t,
-                 void* d_out);
+                 void* d_out, unsigned long long* d_oob);
 
 // CSR (variable-length record) gather: sample `g` owns elements
 // [d_goff[g], d_goff[g+1]) of the global element space; each element is
@@ -56,14 +58,15 @@ void gather_csr(hipStream_t stream,
                 const int64_t* d_idx, int64_t nidx,
                 const int64_t* d_out_off,
                 int64_t elem_bytes,
-                void* d_out);
+                void* d_out, unsigned long long* d_oob);
 
 // Scatter rows of a packed buffer into the local shard at arbitrary local row
 // ids (inverse of gather_rows with nparts==1). Used by the epoch reshuffle to
 // place all-to-all-received rows.
 void scatter_rows_local(hipStream_t stream,
-                        void* d_base, int64_t row_elems, int elem_t,
+                        void* d_base, int64_t nrows_local,
+                        int64_t row_elems, int elem_t,
                         const int64_t* d_local_idx, int64_t nidx,
-                        const void* d_src);
+                        const void* d_src, unsigned long long* d_oob);
 
 } // namespace ddstore

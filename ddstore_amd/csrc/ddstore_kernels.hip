@@ -71,7 +71,7 @@ __device__ __forceinline__ Tout cvt(Tin v) {
 __global__ void __launch_bounds__(kBlock)
 k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int nparts,
                   const int64_t* idx, int64_t nidx, int64_t chunks_per_row,
-                  uint4* __restrict__ out) {
+                  uint4* __restrict__ out, unsigned long long* oob) {
     __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
     __shared__ const uint4* s_base[DDS_MAX_PARTS];
     for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
@@ -85,6 +85,10 @@ k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int npar
         const int64_t r = t / chunks_per_row;
         const int64_t c = t - r * chunks_per_row;
         const int64_t g = idx[r];
+        if (g < 0 || g >= s_prefix[nparts]) {  // skip instead of OOB read
+            if (c == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
         const int p = owner_of(s_prefix, nparts, g);
         out[t] = s_base[p][(g - s_prefix[p]) * chunks_per_row + c];
     }
@@ -106,7 +110,7 @@ template <typename Tin, typename Tout>
 __global__ void __launch_bounds__(kBlock)
 k_gather_rows_castv(const void* const* peer_base, const int64_t* gprefix, int nparts,
                     const int64_t* idx, int64_t nidx, int64_t row_elems,
-                    Tout* __restrict__ out) {
+                    Tout* __restrict__ out, unsigned long long* oob) {
     constexpr int VEC = 16 / (sizeof(Tin) > sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
     using Vin = VecT<Tin, VEC>;
     using Vout = VecT<Tout, VEC>;
@@ -124,6 +128,10 @@ k_gather_rows_castv(const void* const* peer_base, const int64_t* gprefix, int np
         const int64_t r = t / cpr;
         const int64_t c = t - r * cpr;
         const int64_t g = idx[r];
+        if (g < 0 || g >= s_prefix[nparts]) {
+            if (c == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
         const int p = owner_of(s_prefix, nparts, g);
         const Vin vin = *reinterpret_cast<const Vin*>(
             s_base[p] + (g - s_prefix[p]) * row_elems + c * VEC);
@@ -141,7 +149,7 @@ template <typename Tin, typename Tout>
 __global__ void __launch_bounds__(kBlock)
 k_gather_rows_cast(const void* const* peer_base, const int64_t* gprefix, int nparts,
                    const int64_t* idx, int64_t nidx, int64_t row_elems,
-                   Tout* __restrict__ out) {
+                   Tout* __restrict__ out, unsigned long long* oob) {
     __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
     __shared__ const Tin* s_base[DDS_MAX_PARTS];
     for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
@@ -155,6 +163,10 @@ k_gather_rows_cast(const void* const* peer_base, const int64_t* gprefix, int npa
         const int64_t r = t / row_elems;
         const int64_t c = t - r * row_elems;
         const int64_t g = idx[r];
+        if (g < 0 || g >= s_prefix[nparts]) {
+            if (c == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
         const int p = owner_of(s_prefix, nparts, g);
         out[t] = cvt<Tout>(s_base[p][(g - s_prefix[p]) * row_elems + c]);
     }
@@ -173,7 +185,7 @@ k_gather_csr(const void* const* peer_base,
              const int64_t* goff,
              const int64_t* idx, int64_t nidx,
              const int64_t* out_off, int64_t chunks_per_elem,
-             T* __restrict__ out) {
+             T* __restrict__ out, unsigned long long* oob) {
     __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
     __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
     __shared__ const T* s_base[DDS_MAX_PARTS];
@@ -187,6 +199,10 @@ k_gather_csr(const void* const* peer_base,
 
     for (int64_t s = blockIdx.x; s < nidx; s += gridDim.x) {
         const int64_t g = idx[s];
+        if (g < 0 || g >= s_sprefix[nparts]) {
+            if (threadIdx.x == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
         const int p = owner_of(s_sprefix, nparts, g);
         const int64_t e0 = goff[g];
         const int64_t nch = (goff[g + 1] - e0) * chunks_per_elem;
@@ -201,29 +217,39 @@ k_gather_csr(const void* const* peer_base,
 // local_idx[r] of base. Same chunk mapping as gather.
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(kBlock)
-k_scatter_rows_b16(uint4* __restrict__ base, int64_t chunks_per_row,
+k_scatter_rows_b16(uint4* __restrict__ base, int64_t nrows, int64_t chunks_per_row,
                    const int64_t* local_idx, int64_t nidx,
-                   const uint4* __restrict__ src) {
+                   const uint4* __restrict__ src, unsigned long long* oob) {
     const int64_t total = nidx * chunks_per_row;
     for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
          t += (int64_t)gridDim.x * kBlock) {
         const int64_t r = t / chunks_per_row;
         const int64_t c = t - r * chunks_per_row;
-        base[local_idx[r] * chunks_per_row + c] = src[t];
+        const int64_t l = local_idx[r];
+        if (l < 0 || l >= nrows) {
+            if (c == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
+        base[l * chunks_per_row + c] = src[t];
     }
 }
 
 template <typename T>
 __global__ void __launch_bounds__(kBlock)
-k_scatter_rows_elem(T* __restrict__ base, int64_t row_elems,
+k_scatter_rows_elem(T* __restrict__ base, int64_t nrows, int64_t row_elems,
                     const int64_t* local_idx, int64_t nidx,
-                    const T* __restrict__ src) {
+                    const T* __restrict__ src, unsigned long long* oob) {
     const int64_t total = nidx * row_elems;
     for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
          t += (int64_t)gridDim.x * kBlock) {
         const int64_t r = t / row_elems;
         const int64_t c = t - r * row_elems;
-        base[local_idx[r] * row_elems + c] = src[t];
+        const int64_t l = local_idx[r];
+        if (l < 0 || l >= nrows) {
+            if (c == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
+        base[l * row_elems + c] = src[t];
     }
 }
 
@@ -239,27 +265,29 @@ inline int dds_itemsize(int t) {
 template <typename Tin, typename Tout>
 void launch_gather_cast_one(hipStream_t stream, const void* const* pb,
                             const int64_t* pf, int np, const int64_t* idx,
-                            int64_t n, int64_t re, Tout* out) {
+                            int64_t n, int64_t re, Tout* out,
+                            unsigned long long* oob) {
     constexpr int VEC = 16 / (sizeof(Tin) > sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
     if (re % VEC == 0) {
         const int grid = n_blocks(n * (re / VEC));
         hipLaunchKernelGGL((k_gather_rows_castv<Tin, Tout>), dim3(grid), dim3(kBlock),
-                           0, stream, pb, pf, np, idx, n, re, out);
+                           0, stream, pb, pf, np, idx, n, re, out, oob);
     } else {
         const int grid = n_blocks(n * re);
         hipLaunchKernelGGL((k_gather_rows_cast<Tin, Tout>), dim3(grid), dim3(kBlock),
-                           0, stream, pb, pf, np, idx, n, re, out);
+                           0, stream, pb, pf, np, idx, n, re, out, oob);
     }
 }
 
 template <typename Tin>
 void launch_gather_cast_out(hipStream_t stream, const void* const* pb,
                             const int64_t* pf, int np, const int64_t* idx,
-                            int64_t n, int64_t re, int out_t, void* out, int /*grid*/) {
+                            int64_t n, int64_t re, int out_t, void* out,
+                            unsigned long long* oob) {
     switch (out_t) {
 #define DDS_OUT(tag, T)                                                               \
     case tag:                                                                         \
-        launch_gather_cast_one<Tin, T>(stream, pb, pf, np, idx, n, re, (T*)out);      \
+        launch_gather_cast_one<Tin, T>(stream, pb, pf, np, idx, n, re, (T*)out, oob); \
         break;
         DDS_OUT(DDS_U8, uint8_t)
         DDS_OUT(DDS_I32, int32_t)
@@ -279,7 +307,7 @@ void gather_rows(hipStream_t stream,
                  const int64_t* d_prefix, int nparts,
                  const int64_t* d_idx, int64_t nidx,
                  int64_t row_elems, int in_t, int out_t,
-                 void* d_out) {
+                 void* d_out, unsigned long long* d_oob) {
     if (nidx == 0 || row_elems == 0) return;
     const int64_t row_bytes = row_elems * dds_itemsize(in_t);
     if (in_t == out_t && row_bytes % 16 == 0) {
@@ -287,15 +315,14 @@ void gather_rows(hipStream_t stream,
         const int grid = n_blocks(nidx * cpr);
         hipLaunchKernelGGL(k_gather_rows_b16, dim3(grid), dim3(kBlock), 0, stream,
                            d_peer_base, d_prefix, nparts, d_idx, nidx, cpr,
-                           (uint4*)d_out);
+                           (uint4*)d_out, d_oob);
         return;
     }
-    const int grid = n_blocks(nidx * row_elems);
     switch (in_t) {
 #define DDS_IN(tag, T)                                                           \
     case tag:                                                                    \
         launch_gather_cast_out<T>(stream, d_peer_base, d_prefix, nparts, d_idx,  \
-                                  nidx, row_elems, out_t, d_out, grid);          \
+                                  nidx, row_elems, out_t, d_out, d_oob);         \
         break;
         DDS_IN(DDS_U8, uint8_t)
         DDS_IN(DDS_I32, int32_t)
@@ -316,44 +343,46 @@ void gather_csr(hipStream_t stream,
                 const int64_t* d_idx, int64_t nidx,
                 const int64_t* d_out_off,
                 int64_t elem_bytes,
-                void* d_out) {
+                void* d_out, unsigned long long* d_oob) {
     if (nidx == 0) return;
     int grid = (int)(nidx < kMaxBlocks ? nidx : kMaxBlocks);
     if (elem_bytes % 16 == 0) {
         hipLaunchKernelGGL((k_gather_csr<uint4>), dim3(grid), dim3(kBlock), 0, stream,
                            d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
                            d_goff, d_idx, nidx, d_out_off, elem_bytes / 16,
-                           (uint4*)d_out);
+                           (uint4*)d_out, d_oob);
     } else if (elem_bytes % 4 == 0) {
         hipLaunchKernelGGL((k_gather_csr<uint32_t>), dim3(grid), dim3(kBlock), 0, stream,
                            d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
                            d_goff, d_idx, nidx, d_out_off, elem_bytes / 4,
-                           (uint32_t*)d_out);
+                           (uint32_t*)d_out, d_oob);
     } else {
         hipLaunchKernelGGL((k_gather_csr<uint8_t>), dim3(grid), dim3(kBlock), 0, stream,
                            d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
                            d_goff, d_idx, nidx, d_out_off, elem_bytes,
-                           (uint8_t*)d_out);
+                           (uint8_t*)d_out, d_oob);
     }
 }
 
 void scatter_rows_local(hipStream_t stream,
-                        void* d_base, int64_t row_elems, int elem_t,
+                        void* d_base, int64_t nrows_local,
+                        int64_t row_elems, int elem_t,
                         const int64_t* d_local_idx, int64_t nidx,
-                        const void* d_src) {
+                        const void* d_src, unsigned long long* d_oob) {
     if (nidx == 0 || row_elems == 0) return;
     const int64_t row_bytes = row_elems * dds_itemsize(elem_t);
     if (row_bytes % 16 == 0) {
         const int64_t cpr = row_bytes / 16;
         const int grid = n_blocks(nidx * cpr);
         hipLaunchKernelGGL(k_scatter_rows_b16, dim3(grid), dim3(kBlock), 0, stream,
-                           (uint4*)d_base, cpr, d_local_idx, nidx, (const uint4*)d_src);
+                           (uint4*)d_base, nrows_local, cpr, d_local_idx, nidx,
+                           (const uint4*)d_src, d_oob);
     } else {
         const int64_t total_bytes = nidx * row_bytes;
         const int grid = n_blocks(total_bytes);
         hipLaunchKernelGGL((k_scatter_rows_elem<uint8_t>), dim3(grid), dim3(kBlock), 0,
-                           stream, (uint8_t*)d_base, row_bytes, d_local_idx, nidx,
-                           (const uint8_t*)d_src);
+                           stream, (uint8_t*)d_base, nrows_local, row_bytes,
+                           d_local_idx, nidx, (const uint8_t*)d_src, d_oob);
     }
 }
 

@@ -129,8 +129,32 @@ class DDStore:
             self._backend = _C.HostStore(f"{session}g{group_id}", self.rank, self.size)
         self._vars: Dict[str, dict] = {}
         self._freed = False
+        # per-peer traffic accounting (local vs each remote rank), enabled via
+        # DDSTORE_STATS=1 -- the reference has no observability at all
+        # (SURVEY §5); device-side bincount keeps the hot path async.
+        self._stats_enabled = os.environ.get("DDSTORE_STATS", "0") == "1"
 
     # ------------------------------------------------------------------ util
+    def _mkmeta(self, is_csr: bool, dtype, disp: int, counts_all) -> dict:
+        meta = {
+            "is_csr": is_csr,
+            "dtype": dtype,
+            "disp": disp,
+            "nrows_total": sum(counts_all),
+        }
+        if self._stats_enabled:
+            prefix = [0]
+            for c in counts_all:
+                prefix.append(prefix[-1] + int(c))
+            meta["prefix_t"] = torch.tensor(prefix, dtype=torch.int64, device=self.device)
+            meta["peer_rows"] = torch.zeros(self.size, dtype=torch.int64, device=self.device)
+        return meta
+
+    def _account(self, meta: dict, idx: torch.Tensor) -> None:
+        if self._stats_enabled and "prefix_t" in meta:
+            owner = torch.searchsorted(meta["prefix_t"], idx, right=True) - 1
+            meta["peer_rows"] += torch.bincount(owner, minlength=self.size)
+
     def _staged(self, t: torch.Tensor) -> torch.Tensor:
         """Place an input tensor where the backend can ingest it."""
         if self.mode == "shm":
@@ -172,12 +196,7 @@ class DDStore:
         disp = self._validate_uniform(disp, t.dtype)
         nrows_all = self.comm.allgather(nrows)
         self._backend.add(name, self._staged(t), nrows, disp, nrows_all)
-        self._vars[name] = {
-            "is_csr": False,
-            "dtype": t.dtype,
-            "disp": disp,
-            "nrows_total": sum(nrows_all),
-        }
+        self._vars[name] = self._mkmeta(False, t.dtype, disp, nrows_all)
         self._exchange_and_open(name)
 
     def init(
@@ -197,12 +216,7 @@ class DDStore:
         disp = self._validate_uniform(disp, dtype)
         nrows_all = self.comm.allgather(int(nrows))
         self._backend.init(name, int(nrows), disp, torch.empty(0, dtype=dtype).dtype, nrows_all)
-        self._vars[name] = {
-            "is_csr": False,
-            "dtype": dtype,
-            "disp": disp,
-            "nrows_total": sum(nrows_all),
-        }
+        self._vars[name] = self._mkmeta(False, dtype, disp, nrows_all)
         self._exchange_and_open(name)
 
     def update(self, name: str, arr: ArrayLike, offset: int = 0) -> None:
@@ -233,13 +247,8 @@ class DDStore:
             name, self._staged(v), nsamples, nelems, row_elems,
             nsamples_all, nelems_all, goff_t,
         )
-        meta = {
-            "is_csr": True,
-            "dtype": v.dtype,
-            "disp": row_elems,
-            "nrows_total": sum(nsamples_all),
-            "goff": goff_t,
-        }
+        meta = self._mkmeta(True, v.dtype, row_elems, nsamples_all)
+        meta["goff"] = goff_t
         if self.mode == "hip":
             meta["goff_dev"] = goff_t.to(self.device)
         self._vars[name] = meta
@@ -275,6 +284,7 @@ class DDStore:
             out = torch.empty(
                 (n, meta["disp"]), dtype=dtype or meta["dtype"], device=self.device
             )
+        self._account(meta, idx)
         if self.mode == "hip":
             self._backend.gather(name, idx, out)
         else:
@@ -307,6 +317,7 @@ class DDStore:
         total = int(out_off[-1].item())
         if out is None:
             out = torch.empty((total, meta["disp"]), dtype=meta["dtype"], device=self.device)
+        self._account(meta, idx)
         self._backend.gather_csr(name, idx, out_off, out, total)
         return out, out_off
 
@@ -345,6 +356,12 @@ class DDStore:
             out[name] = {
                 k: q[k] for k in ("n_gather", "rows_gathered", "bytes_gathered")
             }
+            meta = self._vars[name]
+            if self._stats_enabled and "peer_rows" in meta:
+                pr = meta["peer_rows"].cpu().tolist()
+                out[name]["rows_by_owner"] = pr
+                out[name]["rows_local"] = pr[self.rank]
+                out[name]["rows_remote"] = sum(pr) - pr[self.rank]
         return out
 
     def _meta(self, name: str) -> dict:

@@ -28,6 +28,7 @@ setup(
                 "ddstore_amd/csrc/ddstore_kernels.hip",
             ],
             include_dirs=[os.path.join(ROOT, "ddstore_amd", "csrc")],
+            libraries=["roctx64"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],

@@ -185,3 +185,20 @@ def test_epoch_fsm_gpu(store):
     with pytest.raises(RuntimeError, match="epoch already began"):
         store.epoch_begin()
     store.epoch_end()
+
+
+def test_oob_index_skipped_and_counted(store):
+    # device gathers must never read out of bounds: bad indices are skipped
+    # and counted (the reference's host path throws; a device tensor of
+    # indices cannot be checked host-side without a sync)
+    arr = torch.randn(32, 8)
+    store.add("ob", arr)
+    idx = torch.tensor([0, 31, 32, -1, 5], dtype=torch.int64)
+    out = store.get_batch("ob", idx)
+    torch.cuda.synchronize()
+    ok = [0, 31, 5]
+    for k, g in enumerate(idx.tolist()):
+        if g in ok:
+            assert torch.equal(out[k].cpu(), arr[g])
+    q = store.query("ob")
+    assert q["oob_skipped"] == 2, q

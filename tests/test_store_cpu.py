@@ -167,3 +167,47 @@ def test_method_param_compat():
         s.get("x", out, 1)
         assert (out == 1).all()
         s.free()
+
+
+def test_free_var_and_readd(store):
+    store.add("tmp", np.ones((4, 2), dtype=np.float32))
+    store._backend.free_var("tmp")
+    del store._vars["tmp"]
+    store.add("tmp", 2 * np.ones((4, 2), dtype=np.float32))
+    out = store.get_batch("tmp", [0])
+    assert out[0, 0].item() == 2.0
+
+
+def test_free_idempotent():
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    s.add("x", np.zeros((2, 2), dtype=np.float32))
+    s.free()
+    s.free()  # must not raise
+
+
+def test_two_stores_coexist():
+    from ddstore_amd import DDStore
+
+    a = DDStore(device="cpu")
+    b = DDStore(device="cpu")
+    a.add("x", np.ones((4, 2), dtype=np.float32))
+    b.add("x", 5 * np.ones((4, 2), dtype=np.float32))
+    assert a.get_batch("x", [0])[0, 0].item() == 1.0
+    assert b.get_batch("x", [0])[0, 0].item() == 5.0
+    a.free()
+    b.free()
+
+
+def test_many_variables(store):
+    for i in range(8):
+        store.add(f"m{i}", np.full((4, 2), float(i), dtype=np.float32))
+    for i in range(8):
+        assert store.get_batch(f"m{i}", [3])[0, 0].item() == float(i)
+
+
+def test_gather_index_out_of_range(store):
+    store.add("x", np.zeros((5, 2), dtype=np.float32))
+    with pytest.raises(RuntimeError, match="out of range"):
+        store.get_batch("x", [5])

@@ -86,15 +86,16 @@ def main():
     total = int(lens.sum())
     vals = torch.randn(total, 1, device=dev)
     s.add_csr("csr", vals, lens)
-    cidx = torch.from_numpy(rng.integers(0, nsamp, size=8192)).to(dev)
+    ncsr = 65536
+    cidx = torch.from_numpy(rng.integers(0, nsamp, size=ncsr)).to(dev)
     goff = torch.from_numpy(np.concatenate([[0], np.cumsum(lens)])).to(dev)
     clens = goff[cidx + 1] - goff[cidx]
-    out_off = torch.zeros(8193, dtype=torch.int64, device=dev)
+    out_off = torch.zeros(ncsr + 1, dtype=torch.int64, device=dev)
     torch.cumsum(clens, 0, out=out_off[1:])
     ctotal = int(out_off[-1])
     cout = torch.empty(ctotal, 1, device=dev)
     el = timeit(lambda: s._backend.gather_csr("csr", cidx, out_off, cout, ctotal))
-    report(f"gather_csr ~1KiB samples (x{8192})", el, ctotal * 8)
+    report(f"gather_csr ~1KiB samples (x{ncsr})", el, ctotal * 8)
 
     s.free()
     if args.json:
