@@ -359,7 +359,8 @@ public:
                             v.d_elem_prefix, nparts_, v.d_goff,
                             idx.data_ptr<int64_t>(), nidx,
                             out_off.data_ptr<int64_t>(),
-                            v.row_elems * v.itemsize, out.data_ptr(), v.d_oob);
+                            v.row_elems * v.itemsize, total_elems,
+                            out.data_ptr(), v.d_oob);
         v.n_gather += 1;
         v.rows_gathered += nidx;
         v.bytes_gathered += total_elems * v.row_elems * v.itemsize;

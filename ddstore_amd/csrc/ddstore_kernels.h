@@ -57,7 +57,7 @@ void gather_csr(hipStream_t stream,
                 const int64_t* d_goff,
                 const int64_t* d_idx, int64_t nidx,
                 const int64_t* d_out_off,
-                int64_t elem_bytes,
+                int64_t elem_bytes, int64_t total_elems,
                 void* d_out, unsigned long long* d_oob);
 
 // Scatter rows of a packed buffer into the local shard at arbitrary local row

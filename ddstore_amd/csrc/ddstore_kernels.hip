@@ -213,6 +213,50 @@ k_gather_csr(const void* const* peer_base,
 }
 
 // ---------------------------------------------------------------------------
+// CSR gather, wave-per-sample variant: for small/medium samples (<= ~4 KiB)
+// a whole 256-thread block per sample leaves most lanes idle after one
+// iteration and caps memory-level parallelism; one 64-lane wave per sample
+// keeps 4x more samples in flight per CU.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void __launch_bounds__(kBlock)
+k_gather_csr_wave(const void* const* peer_base,
+                  const int64_t* sample_prefix, const int64_t* elem_prefix, int nparts,
+                  const int64_t* goff,
+                  const int64_t* idx, int64_t nidx,
+                  const int64_t* out_off, int64_t chunks_per_elem,
+                  T* __restrict__ out, unsigned long long* oob) {
+    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
+    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
+    __shared__ const T* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
+        s_sprefix[i] = sample_prefix[i];
+        s_eprefix[i] = elem_prefix[i];
+    }
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const T*>(peer_base[i]);
+    __syncthreads();
+
+    constexpr int WPB = kBlock / 64;  // waves per block
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    for (int64_t s = (int64_t)blockIdx.x * WPB + wid; s < nidx;
+         s += (int64_t)gridDim.x * WPB) {
+        const int64_t g = idx[s];
+        if (g < 0 || g >= s_sprefix[nparts]) {
+            if (lane == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
+        const int p = owner_of(s_sprefix, nparts, g);
+        const int64_t e0 = goff[g];
+        const int64_t nch = (goff[g + 1] - e0) * chunks_per_elem;
+        const T* src = s_base[p] + (e0 - s_eprefix[p]) * chunks_per_elem;
+        T* dst = out + out_off[s] * chunks_per_elem;
+        for (int64_t c = lane; c < nch; c += 64) dst[c] = src[c];
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Local scatter (reshuffle placement): row r of src -> local row
 // local_idx[r] of base. Same chunk mapping as gather.
 // ---------------------------------------------------------------------------
@@ -342,26 +386,37 @@ void gather_csr(hipStream_t stream,
                 const int64_t* d_goff,
                 const int64_t* d_idx, int64_t nidx,
                 const int64_t* d_out_off,
-                int64_t elem_bytes,
+                int64_t elem_bytes, int64_t total_elems,
                 void* d_out, unsigned long long* d_oob) {
     if (nidx == 0) return;
-    int grid = (int)(nidx < kMaxBlocks ? nidx : kMaxBlocks);
-    if (elem_bytes % 16 == 0) {
-        hipLaunchKernelGGL((k_gather_csr<uint4>), dim3(grid), dim3(kBlock), 0, stream,
-                           d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
-                           d_goff, d_idx, nidx, d_out_off, elem_bytes / 16,
-                           (uint4*)d_out, d_oob);
-    } else if (elem_bytes % 4 == 0) {
-        hipLaunchKernelGGL((k_gather_csr<uint32_t>), dim3(grid), dim3(kBlock), 0, stream,
-                           d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
-                           d_goff, d_idx, nidx, d_out_off, elem_bytes / 4,
-                           (uint32_t*)d_out, d_oob);
+    // wave-per-sample below ~4 KiB average payload, block-per-sample above
+    const int64_t avg_bytes = total_elems > 0 ? total_elems * elem_bytes / nidx : 0;
+    const bool wave = avg_bytes <= 4096;
+    int grid;
+    if (wave) {
+        int64_t b = (nidx + (kBlock / 64) - 1) / (kBlock / 64);
+        grid = (int)(b < kMaxBlocks ? b : kMaxBlocks);
     } else {
-        hipLaunchKernelGGL((k_gather_csr<uint8_t>), dim3(grid), dim3(kBlock), 0, stream,
-                           d_peer_base, d_sample_prefix, d_elem_prefix, nparts,
-                           d_goff, d_idx, nidx, d_out_off, elem_bytes,
-                           (uint8_t*)d_out, d_oob);
+        grid = (int)(nidx < kMaxBlocks ? nidx : kMaxBlocks);
     }
+#define DDS_CSR_LAUNCH(T, div)                                                       \
+    do {                                                                             \
+        if (wave)                                                                    \
+            hipLaunchKernelGGL((k_gather_csr_wave<T>), dim3(grid), dim3(kBlock), 0,  \
+                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,  \
+                               nparts, d_goff, d_idx, nidx, d_out_off,               \
+                               elem_bytes / div, (T*)d_out, d_oob);                  \
+        else                                                                         \
+            hipLaunchKernelGGL((k_gather_csr<T>), dim3(grid), dim3(kBlock), 0,       \
+                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,  \
+                               nparts, d_goff, d_idx, nidx, d_out_off,               \
+                               elem_bytes / div, (T*)d_out, d_oob);                  \
+    } while (0)
+    if (elem_bytes % 16 == 0) DDS_CSR_LAUNCH(uint4, 16);
+    else if (elem_bytes % 8 == 0) DDS_CSR_LAUNCH(uint64_t, 8);
+    else if (elem_bytes % 4 == 0) DDS_CSR_LAUNCH(uint32_t, 4);
+    else DDS_CSR_LAUNCH(uint8_t, 1);
+#undef DDS_CSR_LAUNCH
 }
 
 void scatter_rows_local(hipStream_t stream,

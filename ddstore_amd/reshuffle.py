@@ -2,20 +2,23 @@
 
 New capability vs the reference: the reference only
 shuffles *indices* (DistributedSampler) -- data never moves after ``add``.
-Here the data itself can be redistributed at an epoch boundary with an RCCL
-all-to-all(v), which stripes traffic across all 7 xGMI links per GPU (ring
-collectives would be single-link-bound; pairwise all-to-all reaches the
-~1 TB/s per-GPU aggregate).
+Here the data itself is redistributed at an epoch boundary with an RCCL
+all-to-all(v), which stripes traffic pairwise across all 7 xGMI links per
+GPU (a ring collective would be single-link bound; pairwise all-to-all
+reaches the ~1 TB/s per-GPU aggregate).
 
 The permutation is slot-preserving: after ``reshuffle_epoch(store, name,
-seed)``, global slot ``j`` holds the rows that were previously at slot
+seed)``, global slot ``j`` holds the row that was previously at slot
 ``perm[j]`` -- shard sizes and the prefix directory are unchanged, only the
-contents move. Every rank derives the same permutation from ``seed``, so the
-exchange needs no coordinator.
+contents move. Every rank derives the same permutation from ``seed``
+(``torch.randperm`` with a seeded generator on the store device -- identical
+across ranks of one node, same device architecture), so the exchange needs
+no coordinator. All plan computation (owner lookup, split sizes, placement
+order) runs as torch ops on the store device; the data path is
+gather-kernel -> all_to_all_single -> scatter-kernel.
 """
 from __future__ import annotations
 
-import numpy as np
 import torch
 
 from .store import DDStore
@@ -25,43 +28,50 @@ def reshuffle_epoch(store: DDStore, name: str, seed: int) -> None:
     q = store.query(name)
     if q["is_csr"]:
         raise NotImplementedError("reshuffle of CSR variables is not supported yet")
-    prefix = np.asarray(q["prefix"], dtype=np.int64)
-    ntotal = int(prefix[-1])
+    dev = store.device
+    prefix = torch.tensor(q["prefix"], dtype=torch.int64, device=dev)
+    ntotal = int(q["nrows_total"])
     disp = int(q["disp"])
-    meta = store._meta(name)
-    dtype = meta["dtype"]
+    dtype = store._meta(name)["dtype"]
     size, rank = store.size, store.rank
-    p0, p1 = int(prefix[rank]), int(prefix[rank + 1])
+    p0, p1 = int(q["prefix"][rank]), int(q["prefix"][rank + 1])
 
-    perm = np.random.default_rng(seed).permutation(ntotal)
-    owner_of = lambda j: np.searchsorted(prefix, j, side="right") - 1  # noqa: E731
-    src_of_slot = owner_of(perm)  # rank that currently holds slot j's future row
+    g = torch.Generator(device=dev)
+    g.manual_seed(int(seed))
+    perm = torch.randperm(ntotal, generator=g, device=dev)
+    # rank that currently holds the row destined for slot j
+    src_of_slot = torch.searchsorted(prefix, perm, right=True) - 1
 
-    # --- send side: my rows, grouped by destination rank (j ascending gives
-    # contiguous, ascending dest groups since owner(j) is nondecreasing)
-    j_send = np.nonzero(src_of_slot == rank)[0]
-    dest = owner_of(j_send)
-    send_counts = np.bincount(dest, minlength=size).tolist()
-    send_rows_global = perm[j_send]  # all owned by this rank
-    idx_t = torch.from_numpy(np.ascontiguousarray(send_rows_global))
-    sendbuf = store.get_batch(name, idx_t, dtype=dtype)
+    # --- send side: rows this rank owns, grouped by destination rank
+    # (ascending j gives contiguous, ascending dest groups since owner(j) is
+    # nondecreasing in j)
+    j_send = (src_of_slot == rank).nonzero(as_tuple=True)[0]
+    dest = torch.searchsorted(prefix, j_send, right=True) - 1
+    send_counts = torch.bincount(dest, minlength=size).cpu().tolist()
+    sendbuf = store.get_batch(name, perm[j_send], dtype=dtype)
 
-    # --- recv side: my slots, ordered by (source rank, j) to match the
+    # --- recv side: my slots ordered by (source rank, j) to match the
     # concatenation order all_to_all delivers
     src_mine = src_of_slot[p0:p1]
-    recv_counts = np.bincount(src_mine, minlength=size).tolist()
-    order = np.argsort(src_mine, kind="stable")
-    recvbuf = torch.empty((p1 - p0, disp), dtype=dtype, device=store.device)
+    recv_counts = torch.bincount(src_mine, minlength=size).cpu().tolist()
+    order = torch.argsort(src_mine, stable=True)
+    recvbuf = torch.empty((p1 - p0, disp), dtype=dtype, device=dev)
 
     store.comm.all_to_all_single(
-        recvbuf.view(p1 - p0, disp) if disp else recvbuf,
-        sendbuf.view(len(j_send), disp) if disp else sendbuf,
+        recvbuf,
+        sendbuf.view(-1, disp),
         output_split_sizes=recv_counts,
         input_split_sizes=send_counts,
     )
 
-    dst_local = torch.from_numpy(np.ascontiguousarray(order)).to(store.device)
-    store._backend.scatter_local(name, dst_local, recvbuf)
+    store._backend.scatter_local(name, order.contiguous(), recvbuf)
     if store.mode == "hip":
         torch.cuda.synchronize(store.device)
     store.comm.barrier()
+
+
+def expected_perm(ntotal: int, seed: int, device) -> torch.Tensor:
+    """The permutation a reshuffle with ``seed`` applies (for tests)."""
+    g = torch.Generator(device=device)
+    g.manual_seed(int(seed))
+    return torch.randperm(ntotal, generator=g, device=device)

@@ -130,7 +130,9 @@ def _w_reshuffle(rank, world):
     arr = np.repeat(base[:, None], DIM, axis=1)  # row g == value g
     s.add("x", arr)
     s.reshuffle("x", seed=99)
-    perm = np.random.default_rng(99).permutation(NUM * world)
+    from ddstore_amd.reshuffle import expected_perm
+
+    perm = expected_perm(NUM * world, 99, s.device).numpy()
     out = s.get_batch("x", list(range(NUM * world)))
     assert np.array_equal(out.numpy()[:, 0], perm.astype(np.float32))
     s.free()

@@ -92,10 +92,12 @@ def _w_reshuffle_gpu(rank, world):
     base = torch.arange(rank * NUM, (rank + 1) * NUM, dtype=torch.float32)
     s.add("x", base.unsqueeze(1).repeat(1, DIM))
     s.reshuffle("x", seed=13)
-    perm = np.random.default_rng(13).permutation(NUM * world)
+    from ddstore_amd.reshuffle import expected_perm
+
+    perm = expected_perm(NUM * world, 13, s.device)
     out = s.get_batch("x", list(range(NUM * world)))
     torch.cuda.synchronize()
-    assert torch.equal(out.cpu()[:, 0], torch.from_numpy(perm.astype(np.float32)))
+    assert torch.equal(out.cpu()[:, 0], perm.cpu().to(torch.float32))
     s.free()
 
 

@@ -159,10 +159,12 @@ def test_reshuffle_gpu_single(store):
     arr = torch.arange(256, dtype=torch.float32).repeat_interleave(8).reshape(256, 8)
     store.add("x", arr)
     store.reshuffle("x", seed=11)
-    perm = np.random.default_rng(11).permutation(256)
+    from ddstore_amd.reshuffle import expected_perm
+
+    perm = expected_perm(256, 11, store.device).cpu()
     out = store.get_batch("x", list(range(256)))
     torch.cuda.synchronize()
-    assert torch.equal(out.cpu(), arr[torch.from_numpy(perm)])
+    assert torch.equal(out.cpu(), arr[perm])
 
 
 def test_prefetch_loader_gpu(store):
