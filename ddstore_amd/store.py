@@ -296,6 +296,13 @@ class DDStore:
                 self._backend.gather(name, idx, out)
         return out
 
+    def gather_into(self, name: str, idx_dev: torch.Tensor, out: torch.Tensor) -> None:
+        """Minimal-overhead gather for hot loops: ``idx_dev`` must already be a
+        contiguous int64 tensor on the store device and ``out`` a matching
+        output tensor (as returned/validated by a prior :meth:`get_batch`).
+        Skips Python-side normalization; the native layer still validates."""
+        self._backend.gather(name, idx_dev, out)
+
     def get_csr(
         self,
         name: str,
@@ -344,6 +351,50 @@ class DDStore:
         from .reshuffle import reshuffle_epoch
 
         reshuffle_epoch(self, name, seed)
+
+    # ------------------------------------------------------------ checkpoint
+    def dump(self, name: str, path: str) -> None:
+        """Write this rank's shard (+ metadata) to ``path`` (one file per
+        rank). The reference has no checkpointing; shards there
+        are rebuilt from source data every run."""
+        meta = self._meta(name)
+        q = self._backend.query(name)
+        payload = {
+            "shard": self._backend.local_shard(name).cpu().clone(),
+            "disp": meta["disp"],
+            "dtype": str(meta["dtype"]),
+            "is_csr": meta["is_csr"],
+            "rank": self.rank,
+            "size": self.size,
+            "prefix": list(q["prefix"]),
+        }
+        if meta["is_csr"]:
+            payload["goff"] = meta["goff"]
+            payload["elem_prefix"] = list(q["elem_prefix"])
+        torch.save(payload, path)
+
+    def load(self, name: str, path: str) -> None:
+        """Collective: re-register variable ``name`` from per-rank files
+        written by :meth:`dump` (same world size)."""
+        payload = torch.load(path, weights_only=False)
+        if payload["size"] != self.size:
+            raise ValueError(
+                f"ddstore load: checkpoint world size {payload['size']} != {self.size}"
+            )
+        if payload["rank"] != self.rank:
+            raise ValueError("ddstore load: checkpoint/rank mismatch")
+        shard = payload["shard"]
+        if payload["is_csr"]:
+            goff = payload["goff"]
+            ep = payload["elem_prefix"]
+            p = payload["prefix"]
+            lo, hi = p[self.rank], p[self.rank + 1]
+            lengths = (goff[lo + 1 : hi + 1] - goff[lo:hi]) if hi > lo else torch.zeros(
+                0, dtype=torch.int64
+            )
+            self.add_csr(name, shard, lengths)
+        else:
+            self.add(name, shard)
 
     # ------------------------------------------------------------------- misc
     def query(self, name: str) -> dict:

@@ -213,3 +213,27 @@ def test_gather_index_out_of_range(store):
     store.add("x", np.zeros((5, 2), dtype=np.float32))
     with pytest.raises(RuntimeError, match="out of range"):
         store.get_batch("x", [5])
+
+
+def test_dump_load_roundtrip(tmp_path, store):
+    arr = np.random.rand(16, 4).astype(np.float32)
+    store.add("ck", arr)
+    store.dump("ck", str(tmp_path / "ck.pt"))
+    s2 = DDStore(device="cpu")
+    s2.load("ck", str(tmp_path / "ck.pt"))
+    out = s2.get_batch("ck", list(range(16)))
+    assert np.array_equal(out.numpy(), arr)
+    s2.free()
+
+
+def test_dump_load_csr(tmp_path, store):
+    lengths = [3, 1, 4]
+    vals = np.random.rand(8, 2).astype(np.float64)
+    store.add_csr("ckc", vals, lengths)
+    store.dump("ckc", str(tmp_path / "ckc.pt"))
+    s2 = DDStore(device="cpu")
+    s2.load("ckc", str(tmp_path / "ckc.pt"))
+    v, off = s2.get_csr("ckc", [2, 0, 1])
+    assert off.tolist() == [0, 4, 7, 8]
+    assert np.array_equal(v[:4].numpy(), vals[4:8])
+    s2.free()
