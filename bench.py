@@ -44,11 +44,13 @@ def parse_args():
     p.add_argument("--dim", type=int, default=128, help="f32 elems per row (128 -> 512 B rows)")
     p.add_argument("--batch", type=int, default=131072, help="rows fetched per rank per step")
     p.add_argument("--hidden", type=int, default=1024, help="MLP hidden width")
-    p.add_argument("--mode", choices=["fetch", "train"], default="fetch",
+    p.add_argument("--mode", choices=["fetch", "train", "csr"], default="fetch",
                    help="fetch = global-shuffle sample-fetch throughput, the "
                         "BASELINE.json headline metric (default); train = the "
                         "same fetch overlapped with a full bf16 MLP train step "
-                        "(forward+backward+optimizer, BASELINE config 5)")
+                        "(forward+backward+optimizer, BASELINE config 5); "
+                        "csr = variable-length (HydraGNN-style) sample fetch "
+                        "(BASELINE config 3)")
     p.add_argument("--device", default="cuda")
     return p.parse_args()
 
@@ -95,9 +97,18 @@ def main():
 
     store = DDStore(device=device if use_cuda else "cpu")
     rows, dim, batch = args.rows, args.dim, args.batch
-    shard = torch.randn(rows, dim, dtype=torch.float32,
-                        device=device if use_cuda else "cpu")
-    store.add("bench", shard)
+    tdev = device if use_cuda else "cpu"
+    avg_row_bytes = dim * 4
+    if args.mode == "csr":
+        # variable-length samples: 16..2*dim-16 f32 elements (mean = dim)
+        gcpu = torch.Generator().manual_seed(4321 + rank)
+        lens = torch.randint(16, 2 * dim - 16, (rows,), generator=gcpu)
+        nelems = int(lens.sum())
+        shard = torch.randn(nelems, 1, dtype=torch.float32, device=tdev)
+        store.add_csr("bench", shard, lens)
+    else:
+        shard = torch.randn(rows, dim, dtype=torch.float32, device=tdev)
+        store.add("bench", shard)
     del shard
 
     ntotal = rows * world
@@ -129,7 +140,7 @@ def main():
             for _ in range(n):
                 trainer(next(it))
 
-    else:
+    elif args.mode == "fetch":
         # flagship store metric: back-to-back batched gathers (owner lookup +
         # xGMI peer read + pack + fused bf16 cast per step), nothing else in
         # the timed loop
@@ -160,6 +171,26 @@ def main():
                 k += 1
             counter["k"] = k
 
+    if args.mode == "csr":
+        # variable-length fetch: capacity ring buffers sized for the worst
+        # batch (2*dim elems/sample max), gather_csr per step, no host sync
+        order_dev = order.to(device) if use_cuda else order
+        nring = 4
+        cap = batch * 2 * dim
+        bufs = [torch.empty(cap, 1, dtype=torch.float32, device=tdev) for _ in range(nring)]
+        step_idx = [
+            order_dev[k * batch : (k + 1) * batch].contiguous()
+            for k in range(nsteps_total)
+        ]
+        counter = {"k": 0}
+
+        def run_steps(n: int):
+            k = counter["k"]
+            for _ in range(n):
+                store.get_csr("bench", step_idx[k], out=bufs[k % nring])
+                k += 1
+            counter["k"] = k
+
     run_steps(args.warmup)
 
     if world > 1:
@@ -183,16 +214,16 @@ def main():
         elapsed = float(t.item())
 
     n_samples = world * args.steps * batch
-    row_bytes = dim * 4
+    row_bytes = avg_row_bytes  # csr: mean sample bytes (lengths avg dim elems)
     sps = n_samples / elapsed
     gather_gbps = n_samples * row_bytes / elapsed / 1e9
     remote_gbps = gather_gbps * (world - 1) / world if world > 0 else 0.0
 
     if rank == 0:
         result = {
-            "metric": ("samples/sec (global-shuffle fetch + bf16 train step)"
-                       if args.mode == "train"
-                       else "samples/sec (global-shuffle sample fetch)"),
+            "metric": {"train": "samples/sec (global-shuffle fetch + bf16 train step)",
+                       "fetch": "samples/sec (global-shuffle sample fetch)",
+                       "csr": "samples/sec (global-shuffle variable-length fetch)"}[args.mode],
             "value": sps,
             "unit": "samples/s",
             "n_gpus": world,
@@ -205,8 +236,9 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": f"store-globalshuffle+bf16-mlp{args.hidden}" if args.mode == "train"
-                         else "store-globalshuffle-fetch",
+                "model": {"train": f"store-globalshuffle+bf16-mlp{args.hidden}",
+                          "fetch": "store-globalshuffle-fetch",
+                          "csr": "store-globalshuffle-csr-fetch"}[args.mode],
                 "global_batch": world * batch,
                 "rows_per_rank": rows,
                 "row_bytes": row_bytes,

@@ -321,9 +321,13 @@ class DDStore:
         lens = goff[idx + 1] - goff[idx]
         out_off = torch.zeros(idx.numel() + 1, dtype=torch.int64, device=self.device)
         torch.cumsum(lens, 0, out=out_off[1:])
-        total = int(out_off[-1].item())
         if out is None:
+            total = int(out_off[-1].item())
             out = torch.empty((total, meta["disp"]), dtype=meta["dtype"], device=self.device)
+        else:
+            # caller-provided capacity buffer: no device sync in the hot loop
+            # (the actual packed length is out_off[-1], on device)
+            total = out.numel() // meta["disp"]
         self._account(meta, idx)
         self._backend.gather_csr(name, idx, out_off, out, total)
         return out, out_off
