@@ -52,6 +52,10 @@ def parse_args():
                         "csr = variable-length (HydraGNN-style) sample fetch "
                         "(BASELINE config 3)")
     p.add_argument("--device", default="cuda")
+    p.add_argument("--backend", default=None,
+                   help="torch.distributed backend override (default: nccl on "
+                        "GPU, gloo on CPU; gloo also works on GPU for "
+                        "oversubscribed single-GPU testing)")
     return p.parse_args()
 
 
@@ -85,9 +89,10 @@ def main():
     use_cuda = args.device.startswith("cuda") and torch.cuda.is_available()
 
     if world > 1:
-        backend = "nccl" if use_cuda else "gloo"
+        backend = args.backend or ("nccl" if use_cuda else "gloo")
         dist.init_process_group(backend, rank=rank, world_size=world)
     if use_cuda:
+        local_rank = local_rank % max(torch.cuda.device_count(), 1)
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
     else:
@@ -208,8 +213,8 @@ def main():
 
     elapsed = t1 - t0
     if world > 1:
-        t = torch.tensor([elapsed], device=device if use_cuda else "cpu",
-                         dtype=torch.float64)
+        red_dev = device if (use_cuda and dist.get_backend() == "nccl") else "cpu"
+        t = torch.tensor([elapsed], device=red_dev, dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 

@@ -61,6 +61,10 @@ def main():
     el = timeit(lambda i: s._backend.gather("f32", idxs[i % NIDX], out))
     report("gather f32->f32 (512B rows)", el, B * dim * 8)
 
+    # the obvious PyTorch alternative for the same operation
+    el = timeit(lambda i: torch.index_select(s.local_shard("f32"), 0, idxs[i % NIDX]))
+    report("torch.index_select f32 (baseline)", el, B * dim * 8)
+
     # fused cast f32 -> bf16
     outb = torch.empty(B, dim, device=dev, dtype=torch.bfloat16)
     el = timeit(lambda i: s._backend.gather("f32", idxs[i % NIDX], outb))
