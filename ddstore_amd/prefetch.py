@@ -11,6 +11,7 @@ for sample fetch (BASELINE config 5). Synchronization is via HIP events:
 """
 from __future__ import annotations
 
+import os
 from typing import Iterator, Optional, Sequence, Union
 
 import numpy as np
@@ -45,9 +46,11 @@ class PrefetchLoader:
         out_dtype: Optional[torch.dtype] = None,
         label_name: Optional[str] = None,
         label_dtype: Optional[torch.dtype] = None,
-        depth: int = 2,
+        depth: Optional[int] = None,
         drop_last: bool = False,
     ):
+        if depth is None:
+            depth = int(os.environ.get("DDSTORE_PREFETCH_DEPTH", "2"))
         self.store = store
         self.name = name
         self.label_name = label_name

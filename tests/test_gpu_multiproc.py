@@ -103,3 +103,21 @@ def _w_reshuffle_gpu(rank, world):
 
 def test_reshuffle_gpu_ws2():
     run_dist(_w_reshuffle_gpu, 2)
+
+
+def _w_width_gpu(rank, world):
+    """width groups on GPU: 2 replica groups of 2 ranks, all on cuda:0."""
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cuda:0", ddstore_width=2)
+    assert s.size == 2
+    arr = torch.full((NUM, DIM), float(s.rank + 1))
+    s.add("x", arr)
+    out = s.get_batch("x", [NUM + 3])  # group-rank 1's shard
+    torch.cuda.synchronize()
+    assert out[0, 0].item() == 2.0
+    s.free()
+
+
+def test_width_groups_gpu():
+    run_dist(_w_width_gpu, 4)

@@ -204,3 +204,17 @@ def test_oob_index_skipped_and_counted(store):
             assert torch.equal(out[k].cpu(), arr[g])
     q = store.query("ob")
     assert q["oob_skipped"] == 2, q
+
+
+def test_dump_load_gpu(tmp_path, store):
+    arr = torch.randn(64, 16)
+    store.add("ckg", arr)
+    store.dump("ckg", str(tmp_path / "ckg.pt"))
+    from ddstore_amd import DDStore
+
+    s2 = DDStore(device="cuda:0")
+    s2.load("ckg", str(tmp_path / "ckg.pt"))
+    out = s2.get_batch("ckg", list(range(64)))
+    torch.cuda.synchronize()
+    assert torch.equal(out.cpu(), arr)
+    s2.free()

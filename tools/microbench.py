@@ -65,6 +65,12 @@ def main():
     el = timeit(lambda i: torch.index_select(s.local_shard("f32"), 0, idxs[i % NIDX]))
     report("torch.index_select f32 (baseline)", el, B * dim * 8)
 
+    # sorted indices (same row SET, ascending order): DRAM-locality probe --
+    # legal for training since a shuffled batch is order-invariant
+    sidxs = [torch.sort(i)[0].contiguous() for i in idxs]
+    el = timeit(lambda i: s._backend.gather("f32", sidxs[i % NIDX], out))
+    report("gather f32 512B rows, sorted idx", el, B * dim * 8)
+
     # fused cast f32 -> bf16
     outb = torch.empty(B, dim, device=dev, dtype=torch.bfloat16)
     el = timeit(lambda i: s._backend.gather("f32", idxs[i % NIDX], outb))
