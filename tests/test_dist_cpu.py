@@ -239,3 +239,25 @@ def _w_distdataset(rank, world):
 
 def test_distdataset_ws2():
     run_dist(_w_distdataset, 2)
+
+
+# --------------------------------------------------------------------------
+def _w_empty_shard(rank, world):
+    """A rank with ZERO rows: directory must still route reads correctly
+    (zero-width prefix entries)."""
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    n = 0 if rank == 1 else 8
+    arr = np.full((n, 4), rank + 1.0, dtype=np.float32)
+    s.add("x", arr)
+    q = s.query("x")
+    assert q["nrows_total"] == 8 * (world - 1)
+    out = s.get_batch("x", list(range(q["nrows_total"])))
+    expect = np.repeat([r + 1.0 for r in range(world) if r != 1], 8)
+    assert np.array_equal(out.numpy()[:, 0], expect.astype(np.float32))
+    s.free()
+
+
+def test_empty_shard_rank():
+    run_dist(_w_empty_shard, 3)

@@ -1,0 +1,91 @@
+"""Property-based tests (hypothesis): gather/CSR vs NumPy on random shapes,
+dtypes and index patterns -- the 'golden outputs vs NumPy' strategy from
+SURVEY §4's rebuild test plan, randomized."""
+import numpy as np
+import pytest
+import torch
+from hypothesis import given, settings, strategies as st
+
+from ddstore_amd import DDStore
+
+DTYPES = [np.uint8, np.int32, np.int64, np.float32, np.float64]
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    nrows=st.integers(1, 200),
+    disp=st.integers(1, 40),
+    nbatch=st.integers(0, 300),
+    dt=st.sampled_from(DTYPES),
+    data=st.data(),
+)
+def test_gather_matches_numpy(nrows, disp, nbatch, dt, data):
+    s = DDStore(device="cpu")
+    try:
+        rng = np.random.default_rng(0)
+        arr = (rng.random((nrows, disp)) * 100).astype(dt)
+        s.add("x", arr)
+        idx = data.draw(
+            st.lists(st.integers(0, nrows - 1), min_size=nbatch, max_size=nbatch)
+        )
+        out = s.get_batch("x", np.asarray(idx, dtype=np.int64))
+        assert np.array_equal(out.numpy(), arr[np.asarray(idx, dtype=np.int64)])
+    finally:
+        s.free()
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    lengths=st.lists(st.integers(0, 30), min_size=1, max_size=60),
+    disp=st.integers(1, 8),
+    dt=st.sampled_from([np.float32, np.float64, np.uint8]),
+    data=st.data(),
+)
+def test_csr_matches_numpy(lengths, disp, dt, data):
+    s = DDStore(device="cpu")
+    try:
+        total = sum(lengths)
+        rng = np.random.default_rng(1)
+        vals = (rng.random((max(total, 1), disp)) * 100).astype(dt)[:total]
+        s.add_csr("c", vals.reshape(total, disp), lengths)
+        n = len(lengths)
+        idx = data.draw(st.lists(st.integers(0, n - 1), min_size=0, max_size=80))
+        v, off = s.get_csr("c", np.asarray(idx, dtype=np.int64))
+        goff = np.concatenate([[0], np.cumsum(lengths)])
+        off_l = off.tolist()
+        for k, g in enumerate(idx):
+            seg = v[off_l[k] : off_l[k + 1]].numpy()
+            ref = vals[goff[g] : goff[g + 1]]
+            # an all-empty store cannot infer disp; compare flattened
+            assert np.array_equal(seg.ravel(), ref.ravel())
+    finally:
+        s.free()
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    counts=st.lists(st.integers(0, 30), min_size=1, max_size=5),
+    start_frac=st.floats(0, 1),
+    dt=st.sampled_from([np.float32, np.int64]),
+)
+def test_get_range_single_owner_rule(counts, start_frac, dt):
+    """get() within one shard succeeds; crossing a boundary raises -- same
+    single-owner contract as the reference (ddstore.hpp:210-214), simulated
+    single-rank with a synthetic multi-part directory via per-part adds."""
+    nrows = sum(counts)
+    if nrows == 0:
+        return
+    s = DDStore(device="cpu")
+    try:
+        arr = np.arange(nrows * 2, dtype=dt).reshape(nrows, 2)
+        s.add("x", arr)
+        start = min(int(start_frac * nrows), nrows - 1)
+        count = nrows - start  # always valid single-rank
+        out = np.zeros((count, 2), dtype=dt)
+        s.get("x", out, start=start)
+        assert np.array_equal(out, arr[start : start + count])
+        bad = np.zeros((count + 1, 2), dtype=dt)
+        with pytest.raises(RuntimeError):
+            s.get("x", bad, start=start)
+    finally:
+        s.free()
