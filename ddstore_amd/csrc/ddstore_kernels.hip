@@ -95,42 +95,6 @@ k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int npar
 }
 
 // ---------------------------------------------------------------------------
-// Small-row variant: one THREAD per row. For rows of <=4 chunks the
-// chunk-per-thread mapping loads the row id once per chunk and keeps only
-// one 16-B access in flight per thread; here a thread loads its row id once
-// and issues all of the row's chunk loads back-to-back (independent, so
-// they overlap), then the stores.
-// ---------------------------------------------------------------------------
-template <int CPR>
-__global__ void __launch_bounds__(kBlock)
-k_gather_rows_b16_tpr(const void* const* peer_base, const int64_t* gprefix, int nparts,
-                      const int64_t* idx, int64_t nidx,
-                      uint4* __restrict__ out, unsigned long long* oob) {
-    __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
-    __shared__ const uint4* s_base[DDS_MAX_PARTS];
-    for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
-    for (int i = threadIdx.x; i < nparts; i += kBlock)
-        s_base[i] = reinterpret_cast<const uint4*>(peer_base[i]);
-    __syncthreads();
-
-    for (int64_t r = (int64_t)blockIdx.x * kBlock + threadIdx.x; r < nidx;
-         r += (int64_t)gridDim.x * kBlock) {
-        const int64_t g = idx[r];
-        if (g < 0 || g >= s_prefix[nparts]) {
-            atomicAdd(oob, 1ull);
-            continue;
-        }
-        const int p = owner_of(s_prefix, nparts, g);
-        const uint4* src = s_base[p] + (g - s_prefix[p]) * CPR;
-        uint4 v[CPR];
-#pragma unroll
-        for (int c = 0; c < CPR; ++c) v[c] = src[c];
-#pragma unroll
-        for (int c = 0; c < CPR; ++c) out[r * CPR + c] = v[c];
-    }
-}
-
-// ---------------------------------------------------------------------------
 // Fixed-stride row gather with fused dtype cast, vectorized: each thread
 // moves VEC = 16/max(sizeof(Tin), sizeof(Tout)) elements so the wider side
 // issues full 16-B accesses and the narrower side 16/ratio-B accesses
@@ -392,20 +356,11 @@ void gather_rows(hipStream_t stream,
     const int64_t row_bytes = row_elems * dds_itemsize(in_t);
     if (in_t == out_t && row_bytes % 16 == 0) {
         const int64_t cpr = row_bytes / 16;
-        if (cpr <= 4) {  // small rows: thread-per-row (better ILP per thread)
-            const int grid = n_blocks(nidx);
-            switch (cpr) {
-#define DDS_TPR(N)                                                                   \
-    case N:                                                                          \
-        hipLaunchKernelGGL((k_gather_rows_b16_tpr<N>), dim3(grid), dim3(kBlock), 0,  \
-                           stream, d_peer_base, d_prefix, nparts, d_idx, nidx,       \
-                           (uint4*)d_out, d_oob);                                    \
-        break;
-                DDS_TPR(1) DDS_TPR(2) DDS_TPR(3) DDS_TPR(4)
-#undef DDS_TPR
-            }
-            return;
-        }
+        // NB: a thread-per-row variant (one idx load, all chunks in flight
+        // per thread) was measured SLOWER on MI355X for <=64 B rows (7.5 vs
+        // 6.1 us at B=131072): fewer, fatter threads lose more to the
+        // row-id load+search serialization than they gain in ILP. The
+        // chunk-per-thread mapping below is used for every row size.
         const int grid = n_blocks(nidx * cpr);
         hipLaunchKernelGGL(k_gather_rows_b16, dim3(grid), dim3(kBlock), 0, stream,
                            d_peer_base, d_prefix, nparts, d_idx, nidx, cpr,
