@@ -14,14 +14,12 @@ pyddstore.pyx:61).
 """
 from __future__ import annotations
 
-import math
 from typing import Optional, Sequence
 
 import numpy as np
 import torch
 from torch.utils.data import Dataset
 
-from .comm import as_comm
 from .prefetch import PrefetchLoader
 from .store import DDStore
 

@@ -26,7 +26,7 @@ from __future__ import annotations
 
 import os
 import uuid
-from typing import Dict, List, Optional, Tuple, Union
+from typing import Dict, Optional, Tuple, Union
 
 import numpy as np
 import torch

@@ -3,7 +3,6 @@ dtypes and index patterns -- the 'golden outputs vs NumPy' strategy from
 SURVEY §4's rebuild test plan, randomized."""
 import numpy as np
 import pytest
-import torch
 from hypothesis import given, settings, strategies as st
 
 from ddstore_amd import DDStore
