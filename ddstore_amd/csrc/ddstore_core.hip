@@ -157,26 +157,34 @@ public:
 
     void add(const std::string& name, const at::Tensor& src, int64_t nrows,
              int64_t row_elems, std::vector<int64_t> nrows_all) {
-        DeviceVar& v = new_var(name);
+        check_new(name);
+        TORCH_CHECK(src.is_contiguous(), "ddstore add: array must be C-contiguous");
+        TORCH_CHECK(src.numel() == nrows * row_elems, "ddstore add: shape mismatch");
+        DeviceVar v;
         v.st = src.scalar_type();
         v.dds_t = dds_type_of(src);
         v.itemsize = dds_itemsize(v.dds_t);
         v.row_elems = row_elems;
         v.nrows_local = nrows;
         v.prefix = make_prefix(nrows_all);
-        TORCH_CHECK(src.is_contiguous(), "ddstore add: array must be C-contiguous");
-        TORCH_CHECK(src.numel() == nrows * row_elems, "ddstore add: shape mismatch");
         alloc_base(v, (size_t)(nrows * row_elems * v.itemsize));
-        ingest(v.base, src, 0);
+        try {
+            ingest(v.base, src, 0);
+        } catch (...) {
+            release(v);
+            throw;
+        }
         v.active = true;
+        vars_[name] = std::move(v);
     }
 
     void init(const std::string& name, int64_t nrows, int64_t row_elems,
               at::ScalarType st, std::vector<int64_t> nrows_all) {
         // reference: pre-allocate zeroed shard, fill later via update
         // (ddstore.hpp:110-179; README.md:107 -- no epoch required)
-        DeviceVar& v = new_var(name);
+        check_new(name);
         at::Tensor proto = at::empty({0}, at::TensorOptions().dtype(st));
+        DeviceVar v;
         v.st = st;
         v.dds_t = dds_type_of(proto);
         v.itemsize = dds_itemsize(v.dds_t);
@@ -184,9 +192,15 @@ public:
         v.nrows_local = nrows;
         v.prefix = make_prefix(nrows_all);
         alloc_base(v, (size_t)(nrows * row_elems * v.itemsize));
-        HIP_CHECK(hipMemsetAsync(v.base, 0, v.base_bytes, stream()));
-        HIP_CHECK(hipStreamSynchronize(stream()));
+        try {
+            HIP_CHECK(hipMemsetAsync(v.base, 0, v.base_bytes, stream()));
+            HIP_CHECK(hipStreamSynchronize(stream()));
+        } catch (...) {
+            release(v);
+            throw;
+        }
         v.active = true;
+        vars_[name] = std::move(v);
     }
 
     void add_csr(const std::string& name, const at::Tensor& values,
@@ -197,7 +211,10 @@ public:
         // on an element-addressed store with disp=1 (HydraGNN
         // pattern); here offsets are replicated per GPU and the gather kernel
         // packs whole samples.
-        DeviceVar& v = new_var(name);
+        check_new(name);
+        TORCH_CHECK(values.is_contiguous(), "ddstore add_csr: values must be contiguous");
+        TORCH_CHECK(values.numel() == nelems * row_elems, "ddstore add_csr: shape mismatch");
+        DeviceVar v;
         v.is_csr = true;
         v.st = values.scalar_type();
         v.dds_t = dds_type_of(values);
@@ -207,19 +224,23 @@ public:
         v.nelems_local = nelems;
         v.prefix = make_prefix(nsamples_all);
         v.elem_prefix = make_prefix(nelems_all);
-        TORCH_CHECK(values.is_contiguous(), "ddstore add_csr: values must be contiguous");
-        TORCH_CHECK(values.numel() == nelems * row_elems, "ddstore add_csr: shape mismatch");
         TORCH_CHECK(goff_cpu.scalar_type() == at::kLong && goff_cpu.is_contiguous() &&
                         goff_cpu.device().is_cpu() &&
                         goff_cpu.numel() == v.prefix[nparts_] + 1,
                     "ddstore add_csr: bad global offsets");
         alloc_base(v, (size_t)(nelems * row_elems * v.itemsize));
-        ingest(v.base, values, 0);
-        size_t gbytes = (size_t)goff_cpu.numel() * 8;
-        HIP_CHECK(hipMalloc((void**)&v.d_goff, gbytes));
-        HIP_CHECK(hipMemcpy(v.d_goff, goff_cpu.data_ptr<int64_t>(), gbytes,
-                            hipMemcpyHostToDevice));
+        try {
+            ingest(v.base, values, 0);
+            size_t gbytes = (size_t)goff_cpu.numel() * 8;
+            HIP_CHECK(hipMalloc((void**)&v.d_goff, gbytes));
+            HIP_CHECK(hipMemcpy(v.d_goff, goff_cpu.data_ptr<int64_t>(), gbytes,
+                                hipMemcpyHostToDevice));
+        } catch (...) {
+            release(v);
+            throw;
+        }
         v.active = true;
+        vars_[name] = std::move(v);
     }
 
     py::bytes ipc_handle(const std::string& name) {
@@ -444,9 +465,8 @@ private:
         try { free_all(); } catch (...) { /* after runtime teardown */ }
     }
 
-    DeviceVar& new_var(const std::string& name) {
+    void check_new(const std::string& name) {
         TORCH_CHECK(!vars_.count(name), "ddstore: variable '", name, "' already exists");
-        return vars_[name];
     }
     DeviceVar& var(const std::string& name) {
         auto it = vars_.find(name);
@@ -549,26 +569,30 @@ public:
 
     std::string add(const std::string& name, const at::Tensor& src, int64_t nrows,
                     int64_t row_elems, std::vector<int64_t> nrows_all) {
-        HostVar& v = new_var(name);
+        check_new(name);
+        TORCH_CHECK(src.is_contiguous() && src.device().is_cpu(),
+                    "ddstore add: array must be C-contiguous on CPU");
+        TORCH_CHECK(src.numel() == nrows * row_elems, "ddstore add: shape mismatch");
+        HostVar v;
         v.st = src.scalar_type();
         v.dds_t = dds_type_of(src);
         v.itemsize = dds_itemsize(v.dds_t);
         v.row_elems = row_elems;
         v.nrows_local = nrows;
         v.prefix = make_prefix(nrows_all);
-        TORCH_CHECK(src.is_contiguous() && src.device().is_cpu(),
-                    "ddstore add: array must be C-contiguous on CPU");
-        TORCH_CHECK(src.numel() == nrows * row_elems, "ddstore add: shape mismatch");
         create_shm(v, name, (size_t)(nrows * row_elems * v.itemsize));
         std::memcpy(v.base, src.data_ptr(), (size_t)src.numel() * v.itemsize);
         v.active = true;
-        return v.shm_name;
+        std::string n = v.shm_name;
+        vars_[name] = std::move(v);
+        return n;
     }
 
     std::string init(const std::string& name, int64_t nrows, int64_t row_elems,
                      at::ScalarType st, std::vector<int64_t> nrows_all) {
-        HostVar& v = new_var(name);
+        check_new(name);
         at::Tensor proto = at::empty({0}, at::TensorOptions().dtype(st));
+        HostVar v;
         v.st = st;
         v.dds_t = dds_type_of(proto);
         v.itemsize = dds_itemsize(v.dds_t);
@@ -578,14 +602,17 @@ public:
         create_shm(v, name, (size_t)(nrows * row_elems * v.itemsize));
         std::memset(v.base, 0, v.base_bytes);
         v.active = true;
-        return v.shm_name;
+        std::string n = v.shm_name;
+        vars_[name] = std::move(v);
+        return n;
     }
 
     std::string add_csr(const std::string& name, const at::Tensor& values,
                         int64_t nsamples, int64_t nelems, int64_t row_elems,
                         std::vector<int64_t> nsamples_all,
                         std::vector<int64_t> nelems_all, const at::Tensor& goff_cpu) {
-        HostVar& v = new_var(name);
+        check_new(name);
+        HostVar v;
         v.is_csr = true;
         v.st = values.scalar_type();
         v.dds_t = dds_type_of(values);
@@ -605,7 +632,9 @@ public:
         create_shm(v, name, (size_t)(nelems * row_elems * v.itemsize));
         std::memcpy(v.base, values.data_ptr(), (size_t)values.numel() * v.itemsize);
         v.active = true;
-        return v.shm_name;
+        std::string n = v.shm_name;
+        vars_[name] = std::move(v);
+        return n;
     }
 
     void open_peers(const std::string& name, const std::vector<std::string>& names) {
@@ -802,9 +831,8 @@ public:
     }
 
 private:
-    HostVar& new_var(const std::string& name) {
+    void check_new(const std::string& name) {
         TORCH_CHECK(!vars_.count(name), "ddstore: variable '", name, "' already exists");
-        return vars_[name];
     }
     HostVar& var(const std::string& name) {
         auto it = vars_.find(name);

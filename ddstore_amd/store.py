@@ -260,9 +260,16 @@ class DDStore:
         ``[start, start+len(out))``; the range may not cross a shard boundary
         (reference ddstore.hpp:197-248). One-sided: only this rank
         participates."""
-        t = _as_tensor(out)
-        if isinstance(out, np.ndarray) and t.data_ptr() != torch.from_numpy(out).data_ptr():
-            raise ValueError("ddstore get: output must be C-contiguous")
+        if isinstance(out, np.ndarray):
+            if not out.flags["C_CONTIGUOUS"]:
+                raise ValueError("ddstore get: output must be C-contiguous")
+            t = torch.from_numpy(out)
+        elif isinstance(out, torch.Tensor):
+            if not out.is_contiguous():
+                raise ValueError("ddstore get: output must be C-contiguous")
+            t = out
+        else:
+            raise TypeError("ddstore get: output must be a NumPy array or torch tensor")
         count = int(t.shape[0]) if t.dim() >= 1 else 0
         self._backend.get_range(name, int(start), count, t)
 

@@ -237,3 +237,26 @@ def test_dump_load_csr(tmp_path, store):
     assert off.tolist() == [0, 4, 7, 8]
     assert np.array_equal(v[:4].numpy(), vals[4:8])
     s2.free()
+
+
+def test_get_noncontiguous_output_rejected(store):
+    store.add("x", np.zeros((8, 4), dtype=np.float32))
+    bad = np.zeros((4, 8), dtype=np.float32)[:, ::2]
+    with pytest.raises(ValueError, match="contiguous"):
+        store.get("x", bad, 0)
+    badt = torch.zeros(8, 4)[:, ::2].clone().t()
+    with pytest.raises(ValueError, match="contiguous"):
+        store.get("x", badt.t()[:, ::2] if not badt.is_contiguous() else badt.t(), 0)
+
+
+def test_failed_add_leaves_no_entry(store):
+    bad = np.zeros((4, 8), dtype=np.float32)[:, ::2]  # non-contiguous
+    with pytest.raises((RuntimeError, TypeError, ValueError)):
+        # bypass Python normalization to hit the native contiguity check
+        store._backend.add("ghost", torch.from_numpy(np.zeros((4, 8), dtype=np.float32))[:, ::2], 4, 4, [4])
+    # registration must not have happened
+    with pytest.raises(RuntimeError, match="unknown variable"):
+        store.query("ghost")
+    # and the name is reusable
+    store.add("ghost", np.ones((4, 4), dtype=np.float32))
+    assert store.get_batch("ghost", [0])[0, 0].item() == 1.0
