@@ -111,7 +111,9 @@ __global__ void __launch_bounds__(kBlock)
 k_gather_rows_castv(const void* const* peer_base, const int64_t* gprefix, int nparts,
                     const int64_t* idx, int64_t nidx, int64_t row_elems,
                     Tout* __restrict__ out, unsigned long long* oob) {
-    constexpr int VEC = 16 / (sizeof(Tin) > sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
+    // both sides issue >=16-B accesses: the narrow side exactly 16 B, the
+    // wide side VEC*sizeof 16..128 B (split into dwordx4 by the compiler)
+    constexpr int VEC = 16 / (sizeof(Tin) < sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
     using Vin = VecT<Tin, VEC>;
     using Vout = VecT<Tout, VEC>;
     __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
@@ -311,7 +313,7 @@ void launch_gather_cast_one(hipStream_t stream, const void* const* pb,
                             const int64_t* pf, int np, const int64_t* idx,
                             int64_t n, int64_t re, Tout* out,
                             unsigned long long* oob) {
-    constexpr int VEC = 16 / (sizeof(Tin) > sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
+    constexpr int VEC = 16 / (sizeof(Tin) < sizeof(Tout) ? sizeof(Tin) : sizeof(Tout));
     if (re % VEC == 0) {
         const int grid = n_blocks(n * (re / VEC));
         hipLaunchKernelGGL((k_gather_rows_castv<Tin, Tout>), dim3(grid), dim3(kBlock),
