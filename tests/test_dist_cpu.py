@@ -261,3 +261,24 @@ def _w_empty_shard(rank, world):
 
 def test_empty_shard_rank():
     run_dist(_w_empty_shard, 3)
+
+
+# --------------------------------------------------------------------------
+def _w_reshuffle_width(rank, world):
+    """Reshuffle within replication groups: each width-2 group permutes its
+    own replica independently (split-comm all-to-all)."""
+    from ddstore_amd import DDStore
+    from ddstore_amd.reshuffle import expected_perm
+
+    s = DDStore(device="cpu", ddstore_width=2)
+    base = np.arange(s.rank * NUM, (s.rank + 1) * NUM, dtype=np.float32)
+    s.add("x", np.repeat(base[:, None], DIM, axis=1))
+    s.reshuffle("x", seed=7)
+    perm = expected_perm(2 * NUM, 7, s.device).numpy()
+    out = s.get_batch("x", list(range(2 * NUM)))
+    assert np.array_equal(out.numpy()[:, 0], perm.astype(np.float32))
+    s.free()
+
+
+def test_reshuffle_with_width_groups():
+    run_dist(_w_reshuffle_width, 4)
