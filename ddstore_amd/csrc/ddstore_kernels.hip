@@ -261,6 +261,73 @@ k_gather_csr_wave(const void* const* peer_base,
     }
 }
 
+// Copy nd dwords src->dst by `nthreads` cooperating threads (tid strided),
+// with the STORE side aligned up to dwordx4 (store width is the side the
+// memory system cannot split -- same measured rule as the cast kernels):
+// head dwords to 16-B alignment, uint4 store body fed by dword loads, tail.
+__device__ __forceinline__ void copy_dwords_store16(
+    uint32_t* __restrict__ dst, const uint32_t* __restrict__ src,
+    int64_t nd, int tid, int nthreads) {
+    const int64_t h_align = (int64_t)((4 - ((((uintptr_t)dst) >> 2) & 3)) & 3);
+    const int64_t h = h_align < nd ? h_align : nd;
+    for (int64_t i = tid; i < h; i += nthreads) dst[i] = src[i];
+    const int64_t nb = (nd - h) >> 2;
+    uint4* d4 = reinterpret_cast<uint4*>(dst + h);
+    const uint32_t* s2 = src + h;
+    for (int64_t i = tid; i < nb; i += nthreads) {
+        uint4 v;
+        v.x = s2[4 * i];
+        v.y = s2[4 * i + 1];
+        v.z = s2[4 * i + 2];
+        v.w = s2[4 * i + 3];
+        d4[i] = v;
+    }
+    for (int64_t i = h + (nb << 2) + tid; i < nd; i += nthreads) dst[i] = src[i];
+}
+
+// CSR gather for 4-B-granular (but not 16-B-aligned) elements: dword
+// addressing with 16-B-aligned stores. Wave-per-sample (small) and
+// block-per-sample (large) variants.
+template <bool WAVE>
+__global__ void __launch_bounds__(kBlock)
+k_gather_csr_dw(const void* const* peer_base,
+                const int64_t* sample_prefix, const int64_t* elem_prefix, int nparts,
+                const int64_t* goff,
+                const int64_t* idx, int64_t nidx,
+                const int64_t* out_off, int64_t dwords_per_elem,
+                uint32_t* __restrict__ out, unsigned long long* oob) {
+    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
+    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
+    __shared__ const uint32_t* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
+        s_sprefix[i] = sample_prefix[i];
+        s_eprefix[i] = elem_prefix[i];
+    }
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const uint32_t*>(peer_base[i]);
+    __syncthreads();
+
+    constexpr int WPB = kBlock / 64;
+    const int64_t first = WAVE ? (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6)
+                               : (int64_t)blockIdx.x;
+    const int64_t step = WAVE ? (int64_t)gridDim.x * WPB : (int64_t)gridDim.x;
+    const int tid = WAVE ? (threadIdx.x & 63) : threadIdx.x;
+    const int nthreads = WAVE ? 64 : kBlock;
+    for (int64_t s = first; s < nidx; s += step) {
+        const int64_t g = idx[s];
+        if (g < 0 || g >= s_sprefix[nparts]) {
+            if (tid == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
+        const int p = owner_of(s_sprefix, nparts, g);
+        const int64_t e0 = goff[g];
+        const int64_t nd = (goff[g + 1] - e0) * dwords_per_elem;
+        copy_dwords_store16(out + out_off[s] * dwords_per_elem,
+                            s_base[p] + (e0 - s_eprefix[p]) * dwords_per_elem,
+                            nd, tid, nthreads);
+    }
+}
+
 // ---------------------------------------------------------------------------
 // Local scatter (reshuffle placement): row r of src -> local row
 // local_idx[r] of base. Same chunk mapping as gather.
@@ -422,10 +489,24 @@ void gather_csr(hipStream_t stream,
                                nparts, d_goff, d_idx, nidx, d_out_off,               \
                                elem_bytes / div, (T*)d_out, d_oob);                  \
     } while (0)
-    if (elem_bytes % 16 == 0) DDS_CSR_LAUNCH(uint4, 16);
-    else if (elem_bytes % 8 == 0) DDS_CSR_LAUNCH(uint64_t, 8);
-    else if (elem_bytes % 4 == 0) DDS_CSR_LAUNCH(uint32_t, 4);
-    else DDS_CSR_LAUNCH(uint8_t, 1);
+    if (elem_bytes % 16 == 0) {
+        DDS_CSR_LAUNCH(uint4, 16);
+    } else if (elem_bytes % 4 == 0) {
+        // 4/8-B-granular elements: dword addressing, stores re-aligned to
+        // dwordx4 inside each sample's payload
+        if (wave)
+            hipLaunchKernelGGL((k_gather_csr_dw<true>), dim3(grid), dim3(kBlock), 0,
+                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,
+                               nparts, d_goff, d_idx, nidx, d_out_off,
+                               elem_bytes / 4, (uint32_t*)d_out, d_oob);
+        else
+            hipLaunchKernelGGL((k_gather_csr_dw<false>), dim3(grid), dim3(kBlock), 0,
+                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,
+                               nparts, d_goff, d_idx, nidx, d_out_off,
+                               elem_bytes / 4, (uint32_t*)d_out, d_oob);
+    } else {
+        DDS_CSR_LAUNCH(uint8_t, 1);
+    }
 #undef DDS_CSR_LAUNCH
 }
 

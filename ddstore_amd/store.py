@@ -444,6 +444,12 @@ class DDStore:
         self._vars.clear()
         self._freed = True
 
+    def __reduce__(self):
+        raise TypeError(
+            "DDStore holds GPU/shm state and is not picklable; DataLoader "
+            "workers cannot use it -- use num_workers=0 or PrefetchLoader"
+        )
+
     def __del__(self):
         try:
             if not self._freed:

@@ -218,3 +218,22 @@ def test_dump_load_gpu(tmp_path, store):
     torch.cuda.synchronize()
     assert torch.equal(out.cpu(), arr)
     s2.free()
+
+
+@pytest.mark.parametrize("disp", [1, 2, 4, 7])
+def test_csr_elem_widths_gpu(store, disp):
+    # covers the dwordx4-realigned dword path (4/8 B elems), the uint4 path
+    # (16 B elems) and the byte path (28 B elems? no: 7*4=28 -> dword path).
+    rng = np.random.default_rng(disp)
+    lengths = rng.integers(0, 64, size=300)
+    total = int(lengths.sum())
+    vals = torch.randn(total, disp)
+    store.add_csr(f"cw{disp}", vals, lengths)
+    idx = rng.integers(0, 300, size=128)
+    v, off = store.get_csr(f"cw{disp}", idx)
+    torch.cuda.synchronize()
+    goff = np.concatenate([[0], np.cumsum(lengths)])
+    off_h = off.cpu().tolist()
+    v_h = v.cpu()
+    for k, g in enumerate(idx):
+        assert torch.equal(v_h[off_h[k] : off_h[k + 1]], vals[goff[g] : goff[g + 1]])
