@@ -27,7 +27,7 @@ from .store import DDStore
 def reshuffle_epoch(store: DDStore, name: str, seed: int) -> None:
     q = store.query(name)
     if q["is_csr"]:
-        raise NotImplementedError("reshuffle of CSR variables is not supported yet")
+        return _reshuffle_csr(store, name, seed)
     dev = store.device
     prefix = torch.tensor(q["prefix"], dtype=torch.int64, device=dev)
     ntotal = int(q["nrows_total"])
@@ -68,6 +68,37 @@ def reshuffle_epoch(store: DDStore, name: str, seed: int) -> None:
     if store.mode == "hip":
         torch.cuda.synchronize(store.device)
     store.comm.barrier()
+
+
+def _reshuffle_csr(store: DDStore, name: str, seed: int) -> None:
+    """CSR reshuffle: samples have variable lengths, so per-rank ELEMENT
+    counts change even though sample counts are slot-preserved. Pull-based:
+    every rank one-sided-gathers its new samples (striped across xGMI links
+    by the random permutation), then the variable is re-registered
+    collectively with the new element layout. Transiently holds ~2x the
+    shard (old + new) in memory."""
+    q = store.query(name)
+    meta = store._meta(name)
+    dev = store.device
+    rank = store.rank
+    p0, p1 = int(q["prefix"][rank]), int(q["prefix"][rank + 1])
+    ntotal = int(q["nrows_total"])
+
+    g = torch.Generator(device=dev)
+    g.manual_seed(int(seed))
+    perm = torch.randperm(ntotal, generator=g, device=dev)
+    mine = perm[p0:p1].contiguous()
+
+    values, _ = store.get_csr(name, mine)
+    goff = meta["goff_dev"] if store.mode == "hip" else meta["goff"]
+    lens = (goff[mine + 1] - goff[mine]).cpu()
+    if store.mode == "hip":
+        torch.cuda.synchronize(store.device)
+    store.comm.barrier()  # every rank done pulling before shards are freed
+
+    store._backend.free_var(name)
+    del store._vars[name]
+    store.add_csr(name, values, lens)
 
 
 def expected_perm(ntotal: int, seed: int, device) -> torch.Tensor:

@@ -282,3 +282,40 @@ def _w_reshuffle_width(rank, world):
 
 def test_reshuffle_with_width_groups():
     run_dist(_w_reshuffle_width, 4)
+
+
+# --------------------------------------------------------------------------
+def _w_reshuffle_csr(rank, world):
+    from ddstore_amd import DDStore
+    from ddstore_amd.reshuffle import expected_perm
+
+    s = DDStore(device="cpu")
+    nloc = 20
+    gid0 = rank * nloc
+    rng = np.random.default_rng(rank)
+    lengths = rng.integers(1, 9, size=nloc)
+    vals = np.concatenate(
+        [np.full(l, gid0 + i, dtype=np.float32) for i, l in enumerate(lengths)]
+    ).reshape(-1, 1)
+    s.add_csr("c", vals, lengths)
+    s.reshuffle("c", seed=21)
+    ntotal = nloc * world
+    perm = expected_perm(ntotal, 21, s.device).numpy()
+    v, off = s.get_csr("c", list(range(ntotal)))
+    off = off.tolist()
+    # after reshuffle, slot j must contain OLD sample perm[j]'s payload
+    all_lens = []
+    for r in range(world):
+        rr = np.random.default_rng(r)
+        all_lens.append(rr.integers(1, 9, size=nloc))
+    all_lens = np.concatenate(all_lens)
+    for j in range(ntotal):
+        seg = v[off[j] : off[j + 1], 0].numpy()
+        src = perm[j]
+        assert len(seg) == all_lens[src], (j, src)
+        assert (seg == src).all(), (j, src, seg[:3])
+    s.free()
+
+
+def test_reshuffle_csr_ws3():
+    run_dist(_w_reshuffle_csr, 3)

@@ -237,3 +237,19 @@ def test_csr_elem_widths_gpu(store, disp):
     v_h = v.cpu()
     for k, g in enumerate(idx):
         assert torch.equal(v_h[off_h[k] : off_h[k + 1]], vals[goff[g] : goff[g + 1]])
+
+
+def test_reshuffle_csr_gpu(store):
+    from ddstore_amd.reshuffle import expected_perm
+
+    lengths = [3, 7, 2, 5, 1, 4]
+    vals = torch.cat([torch.full((l, 2), float(i)) for i, l in enumerate(lengths)])
+    store.add_csr("rcg", vals, lengths)
+    store.reshuffle("rcg", seed=31)
+    perm = expected_perm(6, 31, store.device).cpu().tolist()
+    v, off = store.get_csr("rcg", list(range(6)))
+    torch.cuda.synchronize()
+    off = off.cpu().tolist()
+    for j in range(6):
+        seg = v.cpu()[off[j] : off[j + 1]]
+        assert seg.shape[0] == lengths[perm[j]] and (seg == float(perm[j])).all()

@@ -260,3 +260,20 @@ def test_failed_add_leaves_no_entry(store):
     # and the name is reusable
     store.add("ghost", np.ones((4, 4), dtype=np.float32))
     assert store.get_batch("ghost", [0])[0, 0].item() == 1.0
+
+
+def test_reshuffle_csr_single(store):
+    from ddstore_amd.reshuffle import expected_perm
+
+    lengths = [2, 5, 1, 3]
+    vals = np.concatenate(
+        [np.full(l, i, dtype=np.float64) for i, l in enumerate(lengths)]
+    ).reshape(-1, 1)
+    store.add_csr("rc", vals, lengths)
+    store.reshuffle("rc", seed=5)
+    perm = expected_perm(4, 5, store.device).tolist()
+    v, off = store.get_csr("rc", [0, 1, 2, 3])
+    off = off.tolist()
+    for j in range(4):
+        seg = v[off[j] : off[j + 1], 0].numpy()
+        assert len(seg) == lengths[perm[j]] and (seg == perm[j]).all()
