@@ -69,6 +69,8 @@ int dds_type_of(const at::Tensor& t) {
         case at::kDouble: return DDS_F64;
         case at::kHalf: return DDS_F16;
         case at::kBFloat16: return DDS_BF16;
+        case at::kFloat8_e4m3fn: return DDS_F8E4M3;  // OCP fn, matches gfx950
+        case at::kFloat8_e5m2: return DDS_F8E5M2;
         default:
             TORCH_CHECK(false, "ddstore: unsupported dtype ", t.scalar_type());
     }
@@ -76,7 +78,7 @@ int dds_type_of(const at::Tensor& t) {
 
 int64_t dds_itemsize(int t) {
     switch (t) {
-        case DDS_U8: return 1;
+        case DDS_U8: case DDS_F8E4M3: case DDS_F8E5M2: return 1;
         case DDS_F16: case DDS_BF16: return 2;
         case DDS_I32: case DDS_F32: return 4;
         default: return 8;

@@ -20,7 +20,9 @@ enum DDSType : int {
     DDS_F64 = 4,
     DDS_F16 = 5,
     DDS_BF16 = 6,
-    DDS_NUM_TYPES = 7
+    DDS_F8E4M3 = 7,   // OCP e4m3fn (gfx950-native fp8, NOT MI300X fnuz)
+    DDS_F8E5M2 = 8,
+    DDS_NUM_TYPES = 9
 };
 
 // Max world size the in-kernel directory supports (fits in LDS; one node of

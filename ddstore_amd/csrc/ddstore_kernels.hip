@@ -18,6 +18,7 @@
 
 #include <hip/hip_fp16.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <type_traits>
 
 namespace ddstore {
@@ -45,6 +46,10 @@ __device__ __forceinline__ int owner_of(const int64_t* prefix, int nparts, int64
     return lo;
 }
 
+template <typename T>
+constexpr bool is_fp8_v =
+    std::is_same_v<T, __hip_fp8_e4m3> || std::is_same_v<T, __hip_fp8_e5m2>;
+
 template <typename Tout, typename Tin>
 __device__ __forceinline__ Tout cvt(Tin v) {
     if constexpr (std::is_same_v<Tout, Tin>) {
@@ -53,10 +58,14 @@ __device__ __forceinline__ Tout cvt(Tin v) {
         return cvt<Tout, float>(__half2float(v));
     } else if constexpr (std::is_same_v<Tin, __hip_bfloat16>) {
         return cvt<Tout, float>(__bfloat162float(v));
+    } else if constexpr (is_fp8_v<Tin>) {
+        return cvt<Tout, float>(static_cast<float>(v));
     } else if constexpr (std::is_same_v<Tout, __half>) {
         return __float2half(static_cast<float>(v));
     } else if constexpr (std::is_same_v<Tout, __hip_bfloat16>) {
         return __float2bfloat16(static_cast<float>(v));
+    } else if constexpr (is_fp8_v<Tout>) {
+        return Tout(static_cast<float>(v));
     } else {
         return static_cast<Tout>(v);
     }
@@ -412,6 +421,8 @@ void launch_gather_cast_out(hipStream_t stream, const void* const* pb,
         DDS_OUT(DDS_F64, double)
         DDS_OUT(DDS_F16, __half)
         DDS_OUT(DDS_BF16, __hip_bfloat16)
+        DDS_OUT(DDS_F8E4M3, __hip_fp8_e4m3)
+        DDS_OUT(DDS_F8E5M2, __hip_fp8_e5m2)
 #undef DDS_OUT
     }
 }
@@ -452,6 +463,8 @@ void gather_rows(hipStream_t stream,
         DDS_IN(DDS_F64, double)
         DDS_IN(DDS_F16, __half)
         DDS_IN(DDS_BF16, __hip_bfloat16)
+        DDS_IN(DDS_F8E4M3, __hip_fp8_e4m3)
+        DDS_IN(DDS_F8E5M2, __hip_fp8_e5m2)
 #undef DDS_IN
     }
 }

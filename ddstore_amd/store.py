@@ -47,6 +47,8 @@ _SUPPORTED = {
     torch.float64,
     torch.float16,
     torch.bfloat16,
+    torch.float8_e4m3fn,
+    torch.float8_e5m2,
 }
 
 

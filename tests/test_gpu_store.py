@@ -253,3 +253,18 @@ def test_reshuffle_csr_gpu(store):
     for j in range(6):
         seg = v.cpu()[off[j] : off[j + 1]]
         assert seg.shape[0] == lengths[perm[j]] and (seg == float(perm[j])).all()
+
+
+@pytest.mark.parametrize("fp8", [torch.float8_e4m3fn, torch.float8_e5m2])
+def test_fp8_fused_expand_gpu(store, fp8):
+    # store fp8 (half the fetch bytes), gather expanded to bf16/f32 in-kernel;
+    # must match torch's own fp8 decode exactly (OCP e4m3fn/e5m2 on gfx950)
+    arr = torch.randn(512, 32).to(fp8)
+    store.add(f"g8{str(fp8)[-4:]}", arr)
+    idx = torch.randint(0, 512, (128,), dtype=torch.int64)
+    out = store.get_batch(f"g8{str(fp8)[-4:]}", idx, dtype=torch.float32)
+    torch.cuda.synchronize()
+    assert torch.equal(out.cpu(), arr[idx].to(torch.float32))
+    outb = store.get_batch(f"g8{str(fp8)[-4:]}", idx, dtype=torch.bfloat16)
+    torch.cuda.synchronize()
+    assert torch.equal(outb.cpu(), arr[idx].to(torch.bfloat16))

@@ -277,3 +277,15 @@ def test_reshuffle_csr_single(store):
     for j in range(4):
         seg = v[off[j] : off[j + 1], 0].numpy()
         assert len(seg) == lengths[perm[j]] and (seg == perm[j]).all()
+
+
+@pytest.mark.parametrize("fp8", [torch.float8_e4m3fn, torch.float8_e5m2])
+def test_fp8_store_roundtrip(store, fp8):
+    arr = torch.randn(32, 8).to(fp8)
+    store.add(f"f8{str(fp8)[-4:]}", arr)
+    out = store.get_batch(f"f8{str(fp8)[-4:]}", [5, 0, 31])
+    assert out.dtype == fp8
+    assert torch.equal(out.view(torch.uint8), arr[[5, 0, 31]].view(torch.uint8))
+    # CPU cast path (tmp + .to)
+    outf = store.get_batch(f"f8{str(fp8)[-4:]}", [1, 2], dtype=torch.float32)
+    assert torch.equal(outf, arr[[1, 2]].to(torch.float32))
