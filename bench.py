@@ -51,6 +51,9 @@ def parse_args():
                         "(forward+backward+optimizer, BASELINE config 5); "
                         "csr = variable-length (HydraGNN-style) sample fetch "
                         "(BASELINE config 3)")
+    p.add_argument("--store-dtype", choices=["f32", "bf16", "u8", "fp8"], default="f32",
+                   help="shard storage dtype; non-f32 exercises the fused "
+                        "expand-on-gather path (fp8 = OCP e4m3fn)")
     p.add_argument("--device", default="cuda")
     p.add_argument("--backend", default=None,
                    help="torch.distributed backend override (default: nccl on "
@@ -104,6 +107,8 @@ def main():
     rows, dim, batch = args.rows, args.dim, args.batch
     tdev = device if use_cuda else "cpu"
     avg_row_bytes = dim * 4
+    sdt = {"f32": torch.float32, "bf16": torch.bfloat16, "u8": torch.uint8,
+           "fp8": torch.float8_e4m3fn}[args.store_dtype]
     if args.mode == "csr":
         # variable-length samples: 16..2*dim-16 f32 elements (mean = dim)
         gcpu = torch.Generator().manual_seed(4321 + rank)
@@ -113,6 +118,10 @@ def main():
         store.add_csr("bench", shard, lens)
     else:
         shard = torch.randn(rows, dim, dtype=torch.float32, device=tdev)
+        if sdt == torch.uint8:
+            shard = (shard * 64 + 128).clamp(0, 255).to(torch.uint8)
+        elif sdt != torch.float32:
+            shard = shard.to(sdt)
         store.add("bench", shard)
     del shard
 
@@ -176,6 +185,8 @@ def main():
                 k += 1
             counter["k"] = k
 
+    sdt = {"f32": torch.float32, "bf16": torch.bfloat16, "u8": torch.uint8,
+           "fp8": torch.float8_e4m3fn}[args.store_dtype]
     if args.mode == "csr":
         # variable-length fetch: capacity ring buffers sized for the worst
         # batch (2*dim elems/sample max), gather_csr per step, no host sync
@@ -219,7 +230,8 @@ def main():
         elapsed = float(t.item())
 
     n_samples = world * args.steps * batch
-    row_bytes = avg_row_bytes  # csr: mean sample bytes (lengths avg dim elems)
+    stored_itemsize = {"f32": 4, "bf16": 2, "u8": 1, "fp8": 1}[args.store_dtype]
+    row_bytes = dim * (stored_itemsize if args.mode != "csr" else 4)
     sps = n_samples / elapsed
     gather_gbps = n_samples * row_bytes / elapsed / 1e9
     remote_gbps = gather_gbps * (world - 1) / world if world > 0 else 0.0
@@ -247,6 +259,7 @@ def main():
                 "global_batch": world * batch,
                 "rows_per_rank": rows,
                 "row_bytes": row_bytes,
+                "store_dtype": args.store_dtype,
                 "shard_GiB": rows * row_bytes / 2**30,
                 "parallelism": f"dp{world}",
                 "mode": args.mode,

@@ -413,6 +413,10 @@ class DDStore:
     def query(self, name: str) -> dict:
         return self._backend.query(name)
 
+    def variables(self) -> list:
+        """Names of all registered variables."""
+        return list(self._vars)
+
     def stats(self) -> dict:
         out = {}
         for name in list(self._vars):

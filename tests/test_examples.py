@@ -1,0 +1,39 @@
+"""Smoke-run the example scripts (CPU, small sizes) so they never rot."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+pytestmark = pytest.mark.timeout(300)
+
+
+def _run(args):
+    r = subprocess.run([sys.executable] + args, cwd=ROOT, capture_output=True,
+                       text=True, timeout=280)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    return r.stdout
+
+
+def test_demo_example():
+    out = _run(["examples/demo.py", "--num", "5000", "--nbatch", "4"])
+    assert "verified" in out
+
+
+def test_csr_demo_example():
+    out = _run(["examples/csr_demo.py"])
+    assert "verified" in out
+
+
+def test_vae_example():
+    out = _run(["examples/vae_ddp.py", "--epochs", "1", "--nsamples", "1000",
+                "--device", "cpu"])
+    assert "train loss" in out
+
+
+def test_gnn_example():
+    out = _run(["examples/gnn_csr_train.py", "--epochs", "1",
+                "--graphs-per-rank", "1500", "--device", "cpu"])
+    assert "acc" in out
