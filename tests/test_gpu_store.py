@@ -268,3 +268,17 @@ def test_fp8_fused_expand_gpu(store, fp8):
     outb = store.get_batch(f"g8{str(fp8)[-4:]}", idx, dtype=torch.bfloat16)
     torch.cuda.synchronize()
     assert torch.equal(outb.cpu(), arr[idx].to(torch.bfloat16))
+
+
+def test_add_update_from_device_tensors(store):
+    # zero-copy device ingest path (new capability vs the reference's
+    # host-only arrays)
+    arr = torch.randn(64, 8, device="cuda:0")
+    store.add("dev", arr)
+    upd = torch.full((4, 8), 3.5, device="cuda:0")
+    store.update("dev", upd, offset=10)
+    out = store.get_batch("dev", list(range(64)))
+    torch.cuda.synchronize()
+    ref = arr.cpu().clone()
+    ref[10:14] = 3.5
+    assert torch.equal(out.cpu(), ref)
