@@ -20,6 +20,7 @@
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp8.h>
 #include <type_traits>
+#include <cstdlib>
 
 namespace ddstore {
 
@@ -34,6 +35,11 @@ inline int n_blocks(int64_t work_items) {
     if (b > kMaxBlocks) b = kMaxBlocks;
     return (int)b;
 }
+
+template <typename T, int N>
+struct alignas(sizeof(T) * N) VecT {
+    T v[N];
+};
 
 __device__ __forceinline__ int owner_of(const int64_t* prefix, int nparts, int64_t row) {
     // prefix[0] = 0; find p with prefix[p] <= row < prefix[p+1].
@@ -77,17 +83,23 @@ __device__ __forceinline__ Tout cvt(Tin v) {
 // consecutive lanes walk consecutive chunks of one row -> coalesced loads and
 // stores; the per-thread i64 divide is hidden under the memory traffic.
 // ---------------------------------------------------------------------------
+// CHUNK = 16 or 32 bytes per thread. 32 B (two dwordx4 loads in flight per
+// thread) measured faster than 16 B when rows allow -- same finding as the
+// cast kernel, where 32B-read threads beat 16B-read threads on byte rate.
+template <int CHUNK>
 __global__ void __launch_bounds__(kBlock)
 k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int nparts,
                   const int64_t* idx, int64_t nidx, int64_t chunks_per_row,
                   uint4* __restrict__ out, unsigned long long* oob) {
+    using V = VecT<uint4, CHUNK / 16>;
     __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
-    __shared__ const uint4* s_base[DDS_MAX_PARTS];
+    __shared__ const V* s_base[DDS_MAX_PARTS];
     for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
     for (int i = threadIdx.x; i < nparts; i += kBlock)
-        s_base[i] = reinterpret_cast<const uint4*>(peer_base[i]);
+        s_base[i] = reinterpret_cast<const V*>(peer_base[i]);
     __syncthreads();
 
+    V* __restrict__ vout = reinterpret_cast<V*>(out);
     const int64_t total = nidx * chunks_per_row;
     for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
          t += (int64_t)gridDim.x * kBlock) {
@@ -99,7 +111,7 @@ k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int npar
             continue;
         }
         const int p = owner_of(s_prefix, nparts, g);
-        out[t] = s_base[p][(g - s_prefix[p]) * chunks_per_row + c];
+        vout[t] = s_base[p][(g - s_prefix[p]) * chunks_per_row + c];
     }
 }
 
@@ -110,11 +122,6 @@ k_gather_rows_b16(const void* const* peer_base, const int64_t* gprefix, int npar
 // (scalar bf16/f16 loads cost ~2-2.5x -- cdna_hip_programming.md G13).
 // Requires row_elems % VEC == 0 (host falls back to the scalar kernel).
 // ---------------------------------------------------------------------------
-template <typename T, int N>
-struct alignas(sizeof(T) * N) VecT {
-    T v[N];
-};
-
 template <typename Tin, typename Tout>
 __global__ void __launch_bounds__(kBlock)
 k_gather_rows_castv(const void* const* peer_base, const int64_t* gprefix, int nparts,
@@ -443,11 +450,30 @@ void gather_rows(hipStream_t stream,
         // per thread) was measured SLOWER on MI355X for <=64 B rows (7.5 vs
         // 6.1 us at B=131072): fewer, fatter threads lose more to the
         // row-id load+search serialization than they gain in ILP. The
-        // chunk-per-thread mapping below is used for every row size.
-        const int grid = n_blocks(nidx * cpr);
-        hipLaunchKernelGGL(k_gather_rows_b16, dim3(grid), dim3(kBlock), 0, stream,
-                           d_peer_base, d_prefix, nparts, d_idx, nidx, cpr,
-                           (uint4*)d_out, d_oob);
+        // chunk-per-thread mapping is used for every row size, with 32-B
+        // chunks when the row allows (two dwordx4 loads in flight/thread).
+        static const int max_chunk = [] {
+            const char* e = getenv("DDSTORE_CHUNK");
+            return e ? atoi(e) : 32;  // 32 B measured best (see below)
+        }();
+        if (max_chunk >= 64 && row_bytes % 64 == 0) {
+            const int64_t cpr64 = row_bytes / 64;
+            const int grid = n_blocks(nidx * cpr64);
+            hipLaunchKernelGGL((k_gather_rows_b16<64>), dim3(grid), dim3(kBlock), 0,
+                               stream, d_peer_base, d_prefix, nparts, d_idx, nidx,
+                               cpr64, (uint4*)d_out, d_oob);
+        } else if (max_chunk >= 32 && row_bytes % 32 == 0) {
+            const int64_t cpr32 = row_bytes / 32;
+            const int grid = n_blocks(nidx * cpr32);
+            hipLaunchKernelGGL((k_gather_rows_b16<32>), dim3(grid), dim3(kBlock), 0,
+                               stream, d_peer_base, d_prefix, nparts, d_idx, nidx,
+                               cpr32, (uint4*)d_out, d_oob);
+        } else {
+            const int grid = n_blocks(nidx * cpr);
+            hipLaunchKernelGGL((k_gather_rows_b16<16>), dim3(grid), dim3(kBlock), 0,
+                               stream, d_peer_base, d_prefix, nparts, d_idx, nidx,
+                               cpr, (uint4*)d_out, d_oob);
+        }
         return;
     }
     switch (in_t) {
