@@ -106,7 +106,6 @@ def main():
     store = DDStore(device=device if use_cuda else "cpu")
     rows, dim, batch = args.rows, args.dim, args.batch
     tdev = device if use_cuda else "cpu"
-    avg_row_bytes = dim * 4
     sdt = {"f32": torch.float32, "bf16": torch.bfloat16, "u8": torch.uint8,
            "fp8": torch.float8_e4m3fn}[args.store_dtype]
     if args.mode == "csr":
