@@ -452,17 +452,10 @@ void gather_rows(hipStream_t stream,
         // row-id load+search serialization than they gain in ILP. The
         // chunk-per-thread mapping is used for every row size, with 32-B
         // chunks when the row allows (two dwordx4 loads in flight/thread).
-        static const int max_chunk = [] {
-            const char* e = getenv("DDSTORE_CHUNK");
-            return e ? atoi(e) : 32;  // 32 B measured best (see below)
-        }();
-        if (max_chunk >= 64 && row_bytes % 64 == 0) {
-            const int64_t cpr64 = row_bytes / 64;
-            const int grid = n_blocks(nidx * cpr64);
-            hipLaunchKernelGGL((k_gather_rows_b16<64>), dim3(grid), dim3(kBlock), 0,
-                               stream, d_peer_base, d_prefix, nparts, d_idx, nidx,
-                               cpr64, (uint4*)d_out, d_oob);
-        } else if (max_chunk >= 32 && row_bytes % 32 == 0) {
+        // A/B on MI355X (B=131072): 512 B rows -- 16B 30.1us / 32B 28.5 /
+        // 64B 33.9; 64 B rows -- 16B 6.1 / 32B 6.7 (too few threads per
+        // row). 32-B chunks win once a row has >=4 of them.
+        if (row_bytes % 32 == 0 && row_bytes >= 128) {
             const int64_t cpr32 = row_bytes / 32;
             const int grid = n_blocks(nidx * cpr32);
             hipLaunchKernelGGL((k_gather_rows_b16<32>), dim3(grid), dim3(kBlock), 0,
