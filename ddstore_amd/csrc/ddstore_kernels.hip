@@ -348,10 +348,14 @@ k_gather_csr_dw(const void* const* peer_base,
 // Local scatter (reshuffle placement): row r of src -> local row
 // local_idx[r] of base. Same chunk mapping as gather.
 // ---------------------------------------------------------------------------
+template <int CHUNK>
 __global__ void __launch_bounds__(kBlock)
-k_scatter_rows_b16(uint4* __restrict__ base, int64_t nrows, int64_t chunks_per_row,
+k_scatter_rows_b16(uint4* __restrict__ base_, int64_t nrows, int64_t chunks_per_row,
                    const int64_t* local_idx, int64_t nidx,
-                   const uint4* __restrict__ src, unsigned long long* oob) {
+                   const uint4* __restrict__ src_, unsigned long long* oob) {
+    using V = VecT<uint4, CHUNK / 16>;
+    V* __restrict__ base = reinterpret_cast<V*>(base_);
+    const V* __restrict__ src = reinterpret_cast<const V*>(src_);
     const int64_t total = nidx * chunks_per_row;
     for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
          t += (int64_t)gridDim.x * kBlock) {
@@ -549,12 +553,18 @@ void scatter_rows_local(hipStream_t stream,
                         const void* d_src, unsigned long long* d_oob) {
     if (nidx == 0 || row_elems == 0) return;
     const int64_t row_bytes = row_elems * dds_itemsize(elem_t);
-    if (row_bytes % 16 == 0) {
+    if (row_bytes % 32 == 0 && row_bytes >= 128) {
+        const int64_t cpr = row_bytes / 32;
+        const int grid = n_blocks(nidx * cpr);
+        hipLaunchKernelGGL((k_scatter_rows_b16<32>), dim3(grid), dim3(kBlock), 0,
+                           stream, (uint4*)d_base, nrows_local, cpr, d_local_idx,
+                           nidx, (const uint4*)d_src, d_oob);
+    } else if (row_bytes % 16 == 0) {
         const int64_t cpr = row_bytes / 16;
         const int grid = n_blocks(nidx * cpr);
-        hipLaunchKernelGGL(k_scatter_rows_b16, dim3(grid), dim3(kBlock), 0, stream,
-                           (uint4*)d_base, nrows_local, cpr, d_local_idx, nidx,
-                           (const uint4*)d_src, d_oob);
+        hipLaunchKernelGGL((k_scatter_rows_b16<16>), dim3(grid), dim3(kBlock), 0,
+                           stream, (uint4*)d_base, nrows_local, cpr, d_local_idx,
+                           nidx, (const uint4*)d_src, d_oob);
     } else {
         const int64_t total_bytes = nidx * row_bytes;
         const int grid = n_blocks(total_bytes);

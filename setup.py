@@ -20,6 +20,7 @@ setup(
     version="0.1.0",
     description="MI355X-native distributed in-HBM sample store (DDStore capabilities)",
     packages=["ddstore_amd"],
+    py_modules=["pyddstore"],
     ext_modules=[
         CUDAExtension(
             name="ddstore_amd._C",
