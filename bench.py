@@ -159,8 +159,11 @@ def main():
         # the timed loop
         order_dev = order.to(device) if use_cuda else order
         nring = 4
+        # GPU: fused cast to the bf16 training dtype; CPU compat path: the
+        # store dtype (byte move -- the host path has no fused cast)
+        buf_dtype = torch.bfloat16 if use_cuda else sdt
         bufs = [
-            torch.empty(batch, dim, dtype=torch.bfloat16, device=device)
+            torch.empty(batch, dim, dtype=buf_dtype, device=device)
             for _ in range(nring)
         ]
         step_idx = [
