@@ -319,3 +319,20 @@ def _w_reshuffle_csr(rank, world):
 
 def test_reshuffle_csr_ws3():
     run_dist(_w_reshuffle_csr, 3)
+
+
+# --------------------------------------------------------------------------
+def _w_free_after_destroy(rank, world):
+    """free() after torch.distributed teardown must not raise (reference
+    guards MPI_Finalized the same way, ddstore.cxx:81-83)."""
+    from ddstore_amd import DDStore
+
+    s = DDStore(device="cpu")
+    s.add("x", np.ones((8, 2), dtype=np.float32))
+    dist.barrier()
+    dist.destroy_process_group()
+    s.free()  # barrier inside must be skipped gracefully
+
+
+def test_free_after_dist_destroy():
+    run_dist(_w_free_after_destroy, 2)

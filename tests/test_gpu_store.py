@@ -282,3 +282,16 @@ def test_add_update_from_device_tensors(store):
     ref = arr.cpu().clone()
     ref[10:14] = 3.5
     assert torch.equal(out.cpu(), ref)
+
+
+def test_prefetch_with_labels_gpu(store):
+    from ddstore_amd import PrefetchLoader
+
+    data = torch.arange(256, dtype=torch.float32).unsqueeze(1).repeat(1, 8)
+    labels = (torch.arange(256, dtype=torch.int64) * 10).unsqueeze(1)
+    store.add("pd", data)
+    store.add("pl", labels)
+    order = torch.randperm(256)
+    for xb, yb in PrefetchLoader(store, "pd", order, 32, label_name="pl"):
+        torch.cuda.synchronize()
+        assert torch.equal(yb.view(-1).cpu(), (xb[:, 0].cpu() * 10).long())
