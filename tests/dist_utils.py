@@ -30,7 +30,8 @@ def _entry(fn, rank, world, port, backend, args, q):
         try:
             fn(rank, world, *args)
         finally:
-            dist.destroy_process_group()
+            if dist.is_initialized():
+                dist.destroy_process_group()
         q.put((rank, None))
     except Exception:
         q.put((rank, traceback.format_exc()))
