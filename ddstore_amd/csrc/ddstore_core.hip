@@ -363,6 +363,31 @@ public:
         v.bytes_gathered += nidx * v.row_elems * v.itemsize;
     }
 
+    void gather_affine(const std::string& name, const at::Tensor& idx,
+                       at::Tensor out, double scale, double shift) {
+        RoctxRange rr_("ddstore::gather_affine");
+        DeviceVar& v = var(name);
+        TORCH_CHECK(!v.is_csr, "ddstore gather: use gather_csr for CSR variables");
+        check_peers(v);
+        check_idx(idx);
+        int64_t nidx = idx.numel();
+        TORCH_CHECK(out.is_contiguous() && out.device().is_cuda() &&
+                        out.device().index() == device_,
+                    "ddstore gather: output must be a contiguous tensor on the store device");
+        TORCH_CHECK(out.numel() == nidx * v.row_elems, "ddstore gather: shape mismatch");
+        int out_t = dds_type_of(out);
+        TORCH_CHECK(out_t == DDS_F32 || out_t == DDS_F16 || out_t == DDS_BF16,
+                    "ddstore gather: affine output must be f32/f16/bf16");
+        ddstore::gather_rows_affine(stream(), (const void* const*)v.d_peers,
+                                    v.d_prefix, nparts_, idx.data_ptr<int64_t>(),
+                                    nidx, v.row_elems, v.dds_t, out_t,
+                                    (float)scale, (float)shift, out.data_ptr(),
+                                    v.d_oob);
+        v.n_gather += 1;
+        v.rows_gathered += nidx;
+        v.bytes_gathered += nidx * v.row_elems * v.itemsize;
+    }
+
     void gather_csr(const std::string& name, const at::Tensor& idx,
                     const at::Tensor& out_off, at::Tensor out, int64_t total_elems) {
         RoctxRange rr_("ddstore::gather_csr");
@@ -899,6 +924,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("update", &DeviceStore::update)
         .def("get_range", &DeviceStore::get_range)
         .def("gather", &DeviceStore::gather)
+        .def("gather_affine", &DeviceStore::gather_affine)
         .def("gather_csr", &DeviceStore::gather_csr)
         .def("scatter_local", &DeviceStore::scatter_local)
         .def("local_shard", &DeviceStore::local_shard)

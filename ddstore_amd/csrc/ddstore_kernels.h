@@ -46,6 +46,16 @@ void gather_rows(hipStream_t stream,
                  int64_t row_elems, int in_t, int out_t,
                  void* d_out, unsigned long long* d_oob);
 
+// Fused affine gather: out = cast(in) * scale + shift (f32 math; float
+// output dtypes only) -- data-loader normalization folded into the fetch.
+void gather_rows_affine(hipStream_t stream,
+                        const void* const* d_peer_base,
+                        const int64_t* d_prefix, int nparts,
+                        const int64_t* d_idx, int64_t nidx,
+                        int64_t row_elems, int in_t, int out_t,
+                        float scale, float shift,
+                        void* d_out, unsigned long long* d_oob);
+
 // CSR (variable-length record) gather: sample `g` owns elements
 // [d_goff[g], d_goff[g+1]) of the global element space; each element is
 // `elem_bytes` bytes (= disp * itemsize). Output element offsets per sample

@@ -164,6 +164,86 @@ k_gather_rows_castv(const void* const* peer_base, const int64_t* gprefix, int np
 }
 
 // ---------------------------------------------------------------------------
+// Fused affine gather: out = cast(in) * scale + shift, float outputs only
+// (data-loader normalization fused into the fetch: store u8/fp8/f16 samples,
+// gather normalized f32/bf16/f16 minibatches in one pass). Math in f32.
+// ---------------------------------------------------------------------------
+template <typename T>
+__device__ __forceinline__ float to_f32(T v) {
+    if constexpr (std::is_same_v<T, __half>) return __half2float(v);
+    else if constexpr (std::is_same_v<T, __hip_bfloat16>) return __bfloat162float(v);
+    else if constexpr (is_fp8_v<T>) return static_cast<float>(v);
+    else return (float)v;
+}
+
+template <typename Tin, typename Tout>
+__global__ void __launch_bounds__(kBlock)
+k_gather_rows_affine(const void* const* peer_base, const int64_t* gprefix, int nparts,
+                     const int64_t* idx, int64_t nidx, int64_t row_elems,
+                     float scale, float shift,
+                     Tout* __restrict__ out, unsigned long long* oob) {
+    constexpr int VEC = 16 / sizeof(Tout);
+    using Vin = VecT<Tin, VEC>;
+    using Vout = VecT<Tout, VEC>;
+    __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
+    __shared__ const Tin* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const Tin*>(peer_base[i]);
+    __syncthreads();
+
+    const int64_t cpr = row_elems / VEC;
+    const int64_t total = nidx * cpr;
+    for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
+         t += (int64_t)gridDim.x * kBlock) {
+        const int64_t r = t / cpr;
+        const int64_t c = t - r * cpr;
+        const int64_t g = idx[r];
+        if (g < 0 || g >= s_prefix[nparts]) {
+            if (c == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
+        const int p = owner_of(s_prefix, nparts, g);
+        const Vin vin = *reinterpret_cast<const Vin*>(
+            s_base[p] + (g - s_prefix[p]) * row_elems + c * VEC);
+        Vout vout;
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+            vout.v[k] = cvt<Tout>(fmaf(to_f32(vin.v[k]), scale, shift));
+        *reinterpret_cast<Vout*>(out + t * VEC) = vout;
+    }
+}
+
+// element-wise fallback for row_elems not divisible by VEC
+template <typename Tin, typename Tout>
+__global__ void __launch_bounds__(kBlock)
+k_gather_rows_affine_s(const void* const* peer_base, const int64_t* gprefix, int nparts,
+                       const int64_t* idx, int64_t nidx, int64_t row_elems,
+                       float scale, float shift,
+                       Tout* __restrict__ out, unsigned long long* oob) {
+    __shared__ int64_t s_prefix[DDS_MAX_PARTS + 1];
+    __shared__ const Tin* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) s_prefix[i] = gprefix[i];
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const Tin*>(peer_base[i]);
+    __syncthreads();
+    const int64_t total = nidx * row_elems;
+    for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < total;
+         t += (int64_t)gridDim.x * kBlock) {
+        const int64_t r = t / row_elems;
+        const int64_t c = t - r * row_elems;
+        const int64_t g = idx[r];
+        if (g < 0 || g >= s_prefix[nparts]) {
+            if (c == 0) atomicAdd(oob, 1ull);
+            continue;
+        }
+        const int p = owner_of(s_prefix, nparts, g);
+        out[t] = cvt<Tout>(fmaf(
+            to_f32(s_base[p][(g - s_prefix[p]) * row_elems + c]), scale, shift));
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Fixed-stride row gather, general: per-element loop with dtype cast.
 // ---------------------------------------------------------------------------
 template <typename Tin, typename Tout>
@@ -439,6 +519,63 @@ void launch_gather_cast_out(hipStream_t stream, const void* const* pb,
 }
 
 } // namespace
+
+template <typename Tin>
+void launch_affine_out(hipStream_t stream, const void* const* pb, const int64_t* pf,
+                       int np, const int64_t* idx, int64_t n, int64_t re,
+                       float a, float b, int out_t, void* out,
+                       unsigned long long* oob) {
+#define DDS_AFF(tag, T)                                                              \
+    case tag: {                                                                      \
+        constexpr int VEC = 16 / sizeof(T);                                          \
+        if (re % VEC == 0) {                                                         \
+            const int grid = n_blocks(n * (re / VEC));                               \
+            hipLaunchKernelGGL((k_gather_rows_affine<Tin, T>), dim3(grid),           \
+                               dim3(kBlock), 0, stream, pb, pf, np, idx, n, re, a,   \
+                               b, (T*)out, oob);                                     \
+        } else {                                                                     \
+            const int grid = n_blocks(n * re);                                       \
+            hipLaunchKernelGGL((k_gather_rows_affine_s<Tin, T>), dim3(grid),         \
+                               dim3(kBlock), 0, stream, pb, pf, np, idx, n, re, a,   \
+                               b, (T*)out, oob);                                     \
+        }                                                                            \
+        break;                                                                       \
+    }
+    switch (out_t) {
+        DDS_AFF(DDS_F32, float)
+        DDS_AFF(DDS_F16, __half)
+        DDS_AFF(DDS_BF16, __hip_bfloat16)
+    }
+#undef DDS_AFF
+}
+
+void gather_rows_affine(hipStream_t stream,
+                        const void* const* d_peer_base,
+                        const int64_t* d_prefix, int nparts,
+                        const int64_t* d_idx, int64_t nidx,
+                        int64_t row_elems, int in_t, int out_t,
+                        float scale, float shift,
+                        void* d_out, unsigned long long* d_oob) {
+    if (nidx == 0 || row_elems == 0) return;
+    switch (in_t) {
+#define DDS_AIN(tag, T)                                                          \
+    case tag:                                                                    \
+        launch_affine_out<T>(stream, d_peer_base, d_prefix, nparts, d_idx,       \
+                             nidx, row_elems, scale, shift, out_t, d_out,        \
+                             d_oob);                                             \
+        break;
+        DDS_AIN(DDS_U8, uint8_t)
+        DDS_AIN(DDS_I32, int32_t)
+        DDS_AIN(DDS_I64, int64_t)
+        DDS_AIN(DDS_F32, float)
+        DDS_AIN(DDS_F64, double)
+        DDS_AIN(DDS_F16, __half)
+        DDS_AIN(DDS_BF16, __hip_bfloat16)
+        DDS_AIN(DDS_F8E4M3, __hip_fp8_e4m3)
+        DDS_AIN(DDS_F8E5M2, __hip_fp8_e5m2)
+#undef DDS_AIN
+    }
+}
 
 void gather_rows(hipStream_t stream,
                  const void* const* d_peer_base,

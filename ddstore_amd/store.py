@@ -281,10 +281,16 @@ class DDStore:
         indices: ArrayLike,
         out: Optional[torch.Tensor] = None,
         dtype: Optional[torch.dtype] = None,
+        affine: Optional[Tuple[float, float]] = None,
     ) -> torch.Tensor:
         """The hot path: gather an arbitrary batch of global rows in one
         kernel launch (GPU: direct xGMI peer loads), packed (and optionally
-        dtype-cast) into a contiguous ``(n, disp)`` tensor."""
+        dtype-cast) into a contiguous ``(n, disp)`` tensor.
+
+        ``affine=(scale, shift)`` fuses ``out = cast(row) * scale + shift``
+        (f32 math, float output dtypes) into the gather -- data-loader
+        normalization without a second pass (e.g. u8 pixels -> normalized
+        bf16)."""
         meta = self._meta(name)
         idx = torch.as_tensor(indices, dtype=torch.int64)
         idx = idx.to(self.device, non_blocking=True).contiguous()
@@ -294,6 +300,17 @@ class DDStore:
                 (n, meta["disp"]), dtype=dtype or meta["dtype"], device=self.device
             )
         self._account(meta, idx)
+        if affine is not None:
+            scale, shift = float(affine[0]), float(affine[1])
+            if out.dtype not in (torch.float32, torch.float16, torch.bfloat16):
+                raise TypeError("ddstore get_batch: affine output must be float")
+            if self.mode == "hip":
+                self._backend.gather_affine(name, idx, out, scale, shift)
+            else:
+                tmp = torch.empty((n, meta["disp"]), dtype=meta["dtype"])
+                self._backend.gather(name, idx, tmp)
+                out.copy_((tmp.to(torch.float32) * scale + shift).to(out.dtype))
+            return out
         if self.mode == "hip":
             self._backend.gather(name, idx, out)
         else:

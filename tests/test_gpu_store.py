@@ -295,3 +295,26 @@ def test_prefetch_with_labels_gpu(store):
     for xb, yb in PrefetchLoader(store, "pd", order, 32, label_name="pl"):
         torch.cuda.synchronize()
         assert torch.equal(yb.view(-1).cpu(), (xb[:, 0].cpu() * 10).long())
+
+
+def test_affine_gather_gpu(store):
+    # fused normalization: u8 pixels -> normalized f32/bf16, one kernel
+    arr = torch.randint(0, 255, (256, 32), dtype=torch.uint8)
+    store.add("afg", arr)
+    idx = torch.randint(0, 256, (64,), dtype=torch.int64)
+    out = store.get_batch("afg", idx, dtype=torch.float32, affine=(1 / 255.0, -0.5))
+    torch.cuda.synchronize()
+    ref = arr[idx].to(torch.float32) * (1 / 255.0) - 0.5
+    assert torch.allclose(out.cpu(), ref)
+    outb = store.get_batch("afg", idx, dtype=torch.bfloat16, affine=(1 / 255.0, -0.5))
+    torch.cuda.synchronize()
+    refb = (arr[idx].to(torch.float32) * (1 / 255.0) - 0.5).to(torch.bfloat16)
+    # kernel uses fmaf in f32 then casts; allow 1-ulp bf16 differences
+    assert (outb.cpu().to(torch.float32) - refb.to(torch.float32)).abs().max() < 1e-2
+    # f16 store with odd row width -> scalar affine fallback
+    h = torch.randn(100, 5, dtype=torch.float16)
+    store.add("afh", h)
+    out5 = store.get_batch("afh", idx[:32] % 100, dtype=torch.float32, affine=(3.0, 1.0))
+    torch.cuda.synchronize()
+    ref5 = h[(idx[:32] % 100)].to(torch.float32) * 3.0 + 1.0
+    assert torch.allclose(out5.cpu(), ref5, atol=1e-3, rtol=1e-3)

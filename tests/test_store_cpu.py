@@ -289,3 +289,13 @@ def test_fp8_store_roundtrip(store, fp8):
     # CPU cast path (tmp + .to)
     outf = store.get_batch(f"f8{str(fp8)[-4:]}", [1, 2], dtype=torch.float32)
     assert torch.equal(outf, arr[[1, 2]].to(torch.float32))
+
+
+def test_affine_gather_cpu(store):
+    arr = np.random.randint(0, 255, (16, 4)).astype(np.uint8)
+    store.add("af", arr)
+    out = store.get_batch("af", [2, 9], dtype=torch.float32, affine=(2.0, -1.0))
+    ref = torch.from_numpy(arr[[2, 9]]).to(torch.float32) * 2.0 - 1.0
+    assert torch.allclose(out, ref)
+    with pytest.raises(TypeError, match="affine output"):
+        store.get_batch("af", [0], dtype=torch.int32, affine=(1.0, 0.0))
