@@ -231,6 +231,12 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    # integrity: every timed index must have been in range (skipped rows
+    # would mean measuring less work than claimed)
+    qf = store.query("bench")
+    oob = int(qf.get("oob_skipped", 0))
+    assert oob == 0, f"bench integrity: {oob} out-of-range indices were skipped"
+
     n_samples = world * args.steps * batch
     stored_itemsize = {"f32": 4, "bf16": 2, "u8": 1, "fp8": 1}[args.store_dtype]
     row_bytes = dim * (stored_itemsize if args.mode != "csr" else 4)
