@@ -353,8 +353,10 @@ public:
                         out.device().index() == device_,
                     "ddstore gather: output must be a contiguous tensor on the store device");
         TORCH_CHECK(out.numel() == nidx * v.row_elems, "ddstore gather: shape mismatch");
+        // byte move only for the SAME dds type (bool aliases u8); any other
+        // dtype pair converts numerically -- a same-size pair like f16->bf16
+        // must NOT silently reinterpret bits (the CPU path converts too)
         int out_t = dds_type_of(out);
-        if (dds_itemsize(out_t) == v.itemsize && out_t != v.dds_t) out_t = v.dds_t;
         ddstore::gather_rows(stream(), (const void* const*)v.d_peers, v.d_prefix,
                              nparts_, idx.data_ptr<int64_t>(), nidx, v.row_elems,
                              v.dds_t, out_t, out.data_ptr(), v.d_oob);
