@@ -416,6 +416,18 @@ public:
         v.bytes_gathered += total_elems * v.row_elems * v.itemsize;
     }
 
+    void csr_lens(const std::string& name, const at::Tensor& idx, at::Tensor lens) {
+        DeviceVar& v = var(name);
+        TORCH_CHECK(v.is_csr, "ddstore csr_lens: not a CSR variable");
+        check_peers(v);
+        check_idx(idx);
+        TORCH_CHECK(lens.scalar_type() == at::kLong && lens.is_contiguous() &&
+                        lens.device().is_cuda() && lens.numel() == idx.numel(),
+                    "ddstore csr_lens: bad lens tensor");
+        ddstore::csr_lens(stream(), v.d_goff, idx.data_ptr<int64_t>(), idx.numel(),
+                          v.prefix[nparts_], lens.data_ptr<int64_t>(), v.d_oob);
+    }
+
     void scatter_local(const std::string& name, const at::Tensor& local_idx,
                        const at::Tensor& src) {
         // reshuffle placement: src row r -> local row local_idx[r]
@@ -928,6 +940,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("gather", &DeviceStore::gather)
         .def("gather_affine", &DeviceStore::gather_affine)
         .def("gather_csr", &DeviceStore::gather_csr)
+        .def("csr_lens", &DeviceStore::csr_lens)
         .def("scatter_local", &DeviceStore::scatter_local)
         .def("local_shard", &DeviceStore::local_shard)
         .def("epoch_begin", &DeviceStore::epoch_begin)

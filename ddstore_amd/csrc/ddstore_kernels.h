@@ -72,6 +72,11 @@ void gather_csr(hipStream_t stream,
                 int64_t elem_bytes, int64_t total_elems,
                 void* d_out, unsigned long long* d_oob);
 
+// lens[i] = goff[idx[i]+1] - goff[idx[i]] (CSR gather plan helper).
+void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
+              int64_t nidx, int64_t nsamples, int64_t* d_lens,
+              unsigned long long* d_oob);
+
 // Scatter rows of a packed buffer into the local shard at arbitrary local row
 // ids (inverse of gather_rows with nparts==1). Used by the epoch reshuffle to
 // place all-to-all-received rows.

@@ -357,6 +357,24 @@ k_gather_csr_wave(const void* const* peer_base,
     }
 }
 
+// Per-sample lengths for a CSR gather plan: lens[i] = goff[idx[i]+1] -
+// goff[idx[i]] (one kernel instead of the 3-4 elementwise torch launches
+// the equivalent `goff[idx+1]-goff[idx]` costs per step).
+__global__ void __launch_bounds__(kBlock)
+k_csr_lens(const int64_t* goff, const int64_t* idx, int64_t nidx, int64_t nsamples,
+           int64_t* __restrict__ lens, unsigned long long* oob) {
+    for (int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x; i < nidx;
+         i += (int64_t)gridDim.x * kBlock) {
+        const int64_t g = idx[i];
+        if (g < 0 || g >= nsamples) {
+            lens[i] = 0;
+            atomicAdd(oob, 1ull);
+            continue;
+        }
+        lens[i] = goff[g + 1] - goff[g];
+    }
+}
+
 // Copy nd dwords src->dst by `nthreads` cooperating threads (tid strided),
 // with the STORE side aligned up to dwordx4 (store width is the side the
 // memory system cannot split -- same measured rule as the cast kernels):
@@ -627,6 +645,14 @@ void gather_rows(hipStream_t stream,
         DDS_IN(DDS_F8E5M2, __hip_fp8_e5m2)
 #undef DDS_IN
     }
+}
+
+void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
+              int64_t nidx, int64_t nsamples, int64_t* d_lens,
+              unsigned long long* d_oob) {
+    if (nidx == 0) return;
+    hipLaunchKernelGGL(k_csr_lens, dim3(n_blocks(nidx)), dim3(kBlock), 0, stream,
+                       d_goff, d_idx, nidx, nsamples, d_lens, d_oob);
 }
 
 void gather_csr(hipStream_t stream,

@@ -343,9 +343,13 @@ class DDStore:
             raise ValueError(f"ddstore get_csr: '{name}' is not a CSR variable")
         idx = torch.as_tensor(indices, dtype=torch.int64)
         idx = idx.to(self.device, non_blocking=True).contiguous()
-        goff = meta["goff_dev"] if self.mode == "hip" else meta["goff"]
-        lens = goff[idx + 1] - goff[idx]
         out_off = torch.zeros(idx.numel() + 1, dtype=torch.int64, device=self.device)
+        if self.mode == "hip":
+            lens = torch.empty(idx.numel(), dtype=torch.int64, device=self.device)
+            self._backend.csr_lens(name, idx, lens)
+        else:
+            goff = meta["goff"]
+            lens = goff[idx + 1] - goff[idx]
         torch.cumsum(lens, 0, out=out_off[1:])
         if out is None:
             total = int(out_off[-1].item())
