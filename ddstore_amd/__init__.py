@@ -11,6 +11,7 @@ from .store import DDStore
 from .distdataset import DistDataset, nsplit
 from .prefetch import PrefetchLoader
 from .reshuffle import reshuffle_epoch
+from . import io
 
 __version__ = "0.1.0"
 
@@ -22,5 +23,6 @@ __all__ = [
     "nsplit",
     "PrefetchLoader",
     "reshuffle_epoch",
+    "io",
     "__version__",
 ]

@@ -336,3 +336,35 @@ def _w_free_after_destroy(rank, world):
 
 def test_free_after_dist_destroy():
     run_dist(_w_free_after_destroy, 2)
+
+
+# --------------------------------------------------------------------------
+def _w_ingest_files(rank, world, tmpdir):
+    from ddstore_amd import DDStore
+    from ddstore_amd import io as dio
+
+    s = DDStore(device="cpu")
+    dio.add_from_npy(s, "a", f"{tmpdir}/a.npy")
+    full = np.load(f"{tmpdir}/a.npy")
+    out = s.get_batch("a", list(range(full.shape[0])))
+    assert np.array_equal(out.numpy(), full.reshape(full.shape[0], -1))
+    dio.add_from_memmap(s, "b", f"{tmpdir}/b.bin", np.float32, (6,))
+    fullb = np.fromfile(f"{tmpdir}/b.bin", dtype=np.float32).reshape(-1, 6)
+    outb = s.get_batch("b", list(range(fullb.shape[0])))
+    assert np.array_equal(outb.numpy(), fullb)
+    dio.add_csr_from_npy(s, "c", f"{tmpdir}/cv.npy", f"{tmpdir}/cl.npy")
+    lens = np.load(f"{tmpdir}/cl.npy")
+    vals = np.load(f"{tmpdir}/cv.npy")
+    v, off = s.get_csr("c", list(range(len(lens))))
+    assert np.array_equal(v.view(-1).numpy(), vals.reshape(-1))
+    s.free()
+
+
+def test_ingest_from_files(tmp_path):
+    rng = np.random.default_rng(0)
+    np.save(tmp_path / "a.npy", rng.random((50, 2, 3)).astype(np.float32))
+    rng.random((40, 6)).astype(np.float32).tofile(tmp_path / "b.bin")
+    lens = rng.integers(1, 7, size=30)
+    np.save(tmp_path / "cl.npy", lens)
+    np.save(tmp_path / "cv.npy", rng.random((int(lens.sum()), 2)).astype(np.float64))
+    run_dist(_w_ingest_files, 2, str(tmp_path))
