@@ -274,52 +274,12 @@ k_gather_rows_cast(const void* const* peer_base, const int64_t* gprefix, int npa
 }
 
 // ---------------------------------------------------------------------------
-// CSR variable-length gather: one workgroup per sample (grid-stride over
-// samples); within a sample all 256 threads stream the payload with chunked
-// loads (T = uint4 / uint32_t / uint8_t picked host-side from elem_bytes
-// alignment). Samples never cross a shard boundary (enforced at add_csr).
-// ---------------------------------------------------------------------------
-template <typename T>
-__global__ void __launch_bounds__(kBlock)
-k_gather_csr(const void* const* peer_base,
-             const int64_t* sample_prefix, const int64_t* elem_prefix, int nparts,
-             const int64_t* goff,
-             const int64_t* idx, int64_t nidx,
-             const int64_t* out_off, int64_t chunks_per_elem,
-             T* __restrict__ out, unsigned long long* oob) {
-    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
-    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
-    __shared__ const T* s_base[DDS_MAX_PARTS];
-    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
-        s_sprefix[i] = sample_prefix[i];
-        s_eprefix[i] = elem_prefix[i];
-    }
-    for (int i = threadIdx.x; i < nparts; i += kBlock)
-        s_base[i] = reinterpret_cast<const T*>(peer_base[i]);
-    __syncthreads();
-
-    for (int64_t s = blockIdx.x; s < nidx; s += gridDim.x) {
-        const int64_t g = idx[s];
-        if (g < 0 || g >= s_sprefix[nparts]) {
-            if (threadIdx.x == 0) atomicAdd(oob, 1ull);
-            continue;
-        }
-        const int p = owner_of(s_sprefix, nparts, g);
-        const int64_t e0 = goff[g];
-        const int64_t nch = (goff[g + 1] - e0) * chunks_per_elem;
-        const T* src = s_base[p] + (e0 - s_eprefix[p]) * chunks_per_elem;
-        T* dst = out + out_off[s] * chunks_per_elem;
-        for (int64_t c = threadIdx.x; c < nch; c += kBlock) dst[c] = src[c];
-    }
-}
-
-// ---------------------------------------------------------------------------
 // CSR gather, wave-per-sample variant: for small/medium samples (<= ~4 KiB)
 // a whole 256-thread block per sample leaves most lanes idle after one
 // iteration and caps memory-level parallelism; one 64-lane wave per sample
 // keeps 4x more samples in flight per CU.
 // ---------------------------------------------------------------------------
-template <typename T>
+template <typename T, int GROUP>
 __global__ void __launch_bounds__(kBlock)
 k_gather_csr_wave(const void* const* peer_base,
                   const int64_t* sample_prefix, const int64_t* elem_prefix, int nparts,
@@ -338,14 +298,14 @@ k_gather_csr_wave(const void* const* peer_base,
         s_base[i] = reinterpret_cast<const T*>(peer_base[i]);
     __syncthreads();
 
-    constexpr int WPB = kBlock / 64;  // waves per block
-    const int wid = threadIdx.x >> 6;
-    const int lane = threadIdx.x & 63;
-    for (int64_t s = (int64_t)blockIdx.x * WPB + wid; s < nidx;
-         s += (int64_t)gridDim.x * WPB) {
+    constexpr int GPB = kBlock / GROUP;  // sample groups per block
+    const int64_t first = (int64_t)blockIdx.x * GPB + threadIdx.x / GROUP;
+    const int64_t step = (int64_t)gridDim.x * GPB;
+    const int tid = threadIdx.x % GROUP;
+    for (int64_t s = first; s < nidx; s += step) {
         const int64_t g = idx[s];
         if (g < 0 || g >= s_sprefix[nparts]) {
-            if (lane == 0) atomicAdd(oob, 1ull);
+            if (tid == 0) atomicAdd(oob, 1ull);
             continue;
         }
         const int p = owner_of(s_sprefix, nparts, g);
@@ -353,7 +313,7 @@ k_gather_csr_wave(const void* const* peer_base,
         const int64_t nch = (goff[g + 1] - e0) * chunks_per_elem;
         const T* src = s_base[p] + (e0 - s_eprefix[p]) * chunks_per_elem;
         T* dst = out + out_off[s] * chunks_per_elem;
-        for (int64_t c = lane; c < nch; c += 64) dst[c] = src[c];
+        for (int64_t c = tid; c < nch; c += GROUP) dst[c] = src[c];
     }
 }
 
@@ -400,9 +360,10 @@ __device__ __forceinline__ void copy_dwords_store16(
 }
 
 // CSR gather for 4-B-granular (but not 16-B-aligned) elements: dword
-// addressing with 16-B-aligned stores. Wave-per-sample (small) and
-// block-per-sample (large) variants.
-template <bool WAVE>
+// addressing with 16-B-aligned stores. GROUP lanes cooperate per sample
+// (16 for small samples so many samples stay in flight per CU, 64 or the
+// whole block for larger payloads).
+template <int GROUP>
 __global__ void __launch_bounds__(kBlock)
 k_gather_csr_dw(const void* const* peer_base,
                 const int64_t* sample_prefix, const int64_t* elem_prefix, int nparts,
@@ -421,12 +382,10 @@ k_gather_csr_dw(const void* const* peer_base,
         s_base[i] = reinterpret_cast<const uint32_t*>(peer_base[i]);
     __syncthreads();
 
-    constexpr int WPB = kBlock / 64;
-    const int64_t first = WAVE ? (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6)
-                               : (int64_t)blockIdx.x;
-    const int64_t step = WAVE ? (int64_t)gridDim.x * WPB : (int64_t)gridDim.x;
-    const int tid = WAVE ? (threadIdx.x & 63) : threadIdx.x;
-    const int nthreads = WAVE ? 64 : kBlock;
+    constexpr int GPB = kBlock / GROUP;  // sample groups per block
+    const int64_t first = (int64_t)blockIdx.x * GPB + threadIdx.x / GROUP;
+    const int64_t step = (int64_t)gridDim.x * GPB;
+    const int tid = threadIdx.x % GROUP;
     for (int64_t s = first; s < nidx; s += step) {
         const int64_t g = idx[s];
         if (g < 0 || g >= s_sprefix[nparts]) {
@@ -438,7 +397,7 @@ k_gather_csr_dw(const void* const* peer_base,
         const int64_t nd = (goff[g + 1] - e0) * dwords_per_elem;
         copy_dwords_store16(out + out_off[s] * dwords_per_elem,
                             s_base[p] + (e0 - s_eprefix[p]) * dwords_per_elem,
-                            nd, tid, nthreads);
+                            nd, tid, GROUP);
     }
 }
 
@@ -665,50 +624,44 @@ void gather_csr(hipStream_t stream,
                 int64_t elem_bytes, int64_t total_elems,
                 void* d_out, unsigned long long* d_oob) {
     if (nidx == 0) return;
-    // wave-per-sample below ~4 KiB average payload, block-per-sample above
+    // lanes cooperating per sample, picked from the average payload so small
+    // samples keep many transfers in flight per CU (A/B: 131072 x ~512 B
+    // samples ran 77 us wave-per-sample vs 37 us for 2x larger samples)
     const int64_t avg_bytes = total_elems > 0 ? total_elems * elem_bytes / nidx : 0;
-    const bool wave = avg_bytes <= 4096;
-    int grid;
-    if (wave) {
-        int64_t b = (nidx + (kBlock / 64) - 1) / (kBlock / 64);
-        grid = (int)(b < kMaxBlocks ? b : kMaxBlocks);
-    } else {
-        grid = (int)(nidx < kMaxBlocks ? nidx : kMaxBlocks);
-    }
-#define DDS_CSR_LAUNCH(T, div)                                                       \
+    const int group = avg_bytes <= 1024 ? 16 : (avg_bytes <= 8192 ? 64 : 256);
+    int64_t b = (nidx + (kBlock / group) - 1) / (kBlock / group);
+    const int grid = (int)(b < kMaxBlocks ? b : kMaxBlocks);
+#define DDS_CSR_G(T, div, G)                                                         \
+    hipLaunchKernelGGL((k_gather_csr_wave<T, G>), dim3(grid), dim3(kBlock), 0,       \
+                       stream, d_peer_base, d_sample_prefix, d_elem_prefix, nparts,  \
+                       d_goff, d_idx, nidx, d_out_off, elem_bytes / div, (T*)d_out,  \
+                       d_oob)
+#define DDS_CSR_T(T, div)                                                            \
     do {                                                                             \
-        if (wave)                                                                    \
-            hipLaunchKernelGGL((k_gather_csr_wave<T>), dim3(grid), dim3(kBlock), 0,  \
-                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,  \
-                               nparts, d_goff, d_idx, nidx, d_out_off,               \
-                               elem_bytes / div, (T*)d_out, d_oob);                  \
-        else                                                                         \
-            hipLaunchKernelGGL((k_gather_csr<T>), dim3(grid), dim3(kBlock), 0,       \
-                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,  \
-                               nparts, d_goff, d_idx, nidx, d_out_off,               \
-                               elem_bytes / div, (T*)d_out, d_oob);                  \
+        if (group == 16) DDS_CSR_G(T, div, 16);                                      \
+        else if (group == 64) DDS_CSR_G(T, div, 64);                                 \
+        else DDS_CSR_G(T, div, 256);                                                 \
     } while (0)
+#define DDS_CSR_DW(G)                                                                \
+    hipLaunchKernelGGL((k_gather_csr_dw<G>), dim3(grid), dim3(kBlock), 0, stream,    \
+                       d_peer_base, d_sample_prefix, d_elem_prefix, nparts, d_goff,  \
+                       d_idx, nidx, d_out_off, elem_bytes / 4, (uint32_t*)d_out,     \
+                       d_oob)
     if (elem_bytes % 16 == 0) {
-        DDS_CSR_LAUNCH(uint4, 16);
+        DDS_CSR_T(uint4, 16);
     } else if (elem_bytes % 4 == 0) {
         // 4/8-B-granular elements: dword addressing, stores re-aligned to
         // dwordx4 inside each sample's payload
-        if (wave)
-            hipLaunchKernelGGL((k_gather_csr_dw<true>), dim3(grid), dim3(kBlock), 0,
-                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,
-                               nparts, d_goff, d_idx, nidx, d_out_off,
-                               elem_bytes / 4, (uint32_t*)d_out, d_oob);
-        else
-            hipLaunchKernelGGL((k_gather_csr_dw<false>), dim3(grid), dim3(kBlock), 0,
-                               stream, d_peer_base, d_sample_prefix, d_elem_prefix,
-                               nparts, d_goff, d_idx, nidx, d_out_off,
-                               elem_bytes / 4, (uint32_t*)d_out, d_oob);
+        if (group == 16) DDS_CSR_DW(16);
+        else if (group == 64) DDS_CSR_DW(64);
+        else DDS_CSR_DW(256);
     } else {
-        DDS_CSR_LAUNCH(uint8_t, 1);
+        DDS_CSR_T(uint8_t, 1);
     }
-#undef DDS_CSR_LAUNCH
+#undef DDS_CSR_G
+#undef DDS_CSR_T
+#undef DDS_CSR_DW
 }
-
 void scatter_rows_local(hipStream_t stream,
                         void* d_base, int64_t nrows_local,
                         int64_t row_elems, int elem_t,
