@@ -624,11 +624,19 @@ void gather_csr(hipStream_t stream,
                 int64_t elem_bytes, int64_t total_elems,
                 void* d_out, unsigned long long* d_oob) {
     if (nidx == 0) return;
-    // lanes cooperating per sample, picked from the average payload so small
-    // samples keep many transfers in flight per CU (A/B: 131072 x ~512 B
-    // samples ran 77 us wave-per-sample vs 37 us for 2x larger samples)
-    const int64_t avg_bytes = total_elems > 0 ? total_elems * elem_bytes / nidx : 0;
-    const int group = avg_bytes <= 1024 ? 16 : (avg_bytes <= 8192 ? 64 : 256);
+    // lanes cooperating per sample: small groups amortize the per-sample
+    // setup (row-id load + directory search) across the wave and keep more
+    // samples in flight. A/B on MI355X: GROUP=16 wins or ties from 512 B to
+    // 32 KiB average samples (2.23 vs 1.44 G samples/s at 512 B vs GROUP=64)
+    // -- the only reason to grow the group is chip COVERAGE when the batch
+    // has few samples (need >=1024 workgroups across 256 CUs).
+    (void)total_elems;
+    static const int g_override = [] {
+        const char* e = getenv("DDSTORE_CSR_GROUP");
+        return e ? atoi(e) : 0;
+    }();
+    int group = nidx >= 16384 ? 16 : (nidx >= 4096 ? 64 : 256);
+    if (g_override) group = g_override;
     int64_t b = (nidx + (kBlock / group) - 1) / (kBlock / group);
     const int grid = (int)(b < kMaxBlocks ? b : kMaxBlocks);
 #define DDS_CSR_G(T, div, G)                                                         \
@@ -638,7 +646,8 @@ void gather_csr(hipStream_t stream,
                        d_oob)
 #define DDS_CSR_T(T, div)                                                            \
     do {                                                                             \
-        if (group == 16) DDS_CSR_G(T, div, 16);                                      \
+        if (group == 8) DDS_CSR_G(T, div, 8);                                        \
+        else if (group == 16) DDS_CSR_G(T, div, 16);                                  \
         else if (group == 64) DDS_CSR_G(T, div, 64);                                 \
         else DDS_CSR_G(T, div, 256);                                                 \
     } while (0)
@@ -652,7 +661,8 @@ void gather_csr(hipStream_t stream,
     } else if (elem_bytes % 4 == 0) {
         // 4/8-B-granular elements: dword addressing, stores re-aligned to
         // dwordx4 inside each sample's payload
-        if (group == 16) DDS_CSR_DW(16);
+        if (group == 8) DDS_CSR_DW(8);
+        else if (group == 16) DDS_CSR_DW(16);
         else if (group == 64) DDS_CSR_DW(64);
         else DDS_CSR_DW(256);
     } else {
