@@ -274,10 +274,8 @@ k_gather_rows_cast(const void* const* peer_base, const int64_t* gprefix, int npa
 }
 
 // ---------------------------------------------------------------------------
-// CSR gather, wave-per-sample variant: for small/medium samples (<= ~4 KiB)
-// a whole 256-thread block per sample leaves most lanes idle after one
-// iteration and caps memory-level parallelism; one 64-lane wave per sample
-// keeps 4x more samples in flight per CU.
+// CSR gather, chunked: GROUP lanes cooperate per sample (see gather_csr's
+// heuristic note; small groups amortize per-sample setup across the wave).
 // ---------------------------------------------------------------------------
 template <typename T, int GROUP>
 __global__ void __launch_bounds__(kBlock)
@@ -448,7 +446,7 @@ k_scatter_rows_elem(T* __restrict__ base, int64_t nrows, int64_t row_elems,
 
 inline int dds_itemsize(int t) {
     switch (t) {
-        case DDS_U8: return 1;
+        case DDS_U8: case DDS_F8E4M3: case DDS_F8E5M2: return 1;
         case DDS_F16: case DDS_BF16: return 2;
         case DDS_I32: case DDS_F32: return 4;
         default: return 8;

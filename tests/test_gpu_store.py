@@ -38,7 +38,7 @@ def test_add_gather_roundtrip(store):
 @pytest.mark.parametrize(
     "dtype",
     [torch.uint8, torch.int32, torch.int64, torch.float32, torch.float64,
-     torch.float16, torch.bfloat16],
+     torch.float16, torch.bfloat16, torch.float8_e4m3fn, torch.float8_e5m2],
 )
 def test_gather_dtypes(store, dtype):
     if dtype.is_floating_point:
@@ -49,7 +49,7 @@ def test_gather_dtypes(store, dtype):
     idx = torch.randint(0, 128, (64,), dtype=torch.int64)
     out = store.get_batch(f"v{str(dtype)}", idx)
     torch.cuda.synchronize()
-    assert torch.equal(out.cpu(), arr[idx])
+    assert torch.equal(out.cpu().view(torch.uint8), arr[idx].view(torch.uint8))
 
 
 def test_gather_odd_row_bytes(store):
