@@ -318,3 +318,19 @@ def test_affine_gather_gpu(store):
     torch.cuda.synchronize()
     ref5 = h[(idx[:32] % 100)].to(torch.float32) * 3.0 + 1.0
     assert torch.allclose(out5.cpu(), ref5, atol=1e-3, rtol=1e-3)
+
+
+@pytest.mark.parametrize("dtype,dim", [(torch.float16, 4), (torch.float32, 3),
+                                        (torch.float8_e4m3fn, 16)])
+def test_reshuffle_odd_rows_gpu(store, dtype, dim):
+    # exercises the scatter byte path (row_bytes % 16 != 0) and 1-byte dtypes
+    from ddstore_amd.reshuffle import expected_perm
+
+    n = 128
+    arr = (torch.arange(n, dtype=torch.float32) % 13).unsqueeze(1).repeat(1, dim).to(dtype)
+    store.add(f"ro{dim}{str(dtype)[-4:]}", arr)
+    store.reshuffle(f"ro{dim}{str(dtype)[-4:]}", seed=77)
+    perm = expected_perm(n, 77, store.device).cpu()
+    out = store.get_batch(f"ro{dim}{str(dtype)[-4:]}", list(range(n)))
+    torch.cuda.synchronize()
+    assert torch.equal(out.cpu().view(torch.uint8), arr[perm].view(torch.uint8))
