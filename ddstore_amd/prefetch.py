@@ -61,6 +61,13 @@ class PrefetchLoader:
         self.drop_last = drop_last
         idx = torch.as_tensor(indices, dtype=torch.int64).flatten()
         self.indices = idx
+        self.is_csr = store._meta(name)["is_csr"]
+        if self.is_csr:
+            if label_name is not None and store._meta(label_name)["is_csr"]:
+                raise ValueError("label variable must be fixed-stride")
+            # capacity ring buffers sized for the worst batch
+            goff = store._meta(name)["goff"]
+            self._max_len = int((goff[1:] - goff[:-1]).max().item()) if goff.numel() > 1 else 0
 
     def __len__(self) -> int:
         n = self.indices.numel()
@@ -71,14 +78,36 @@ class PrefetchLoader:
         stop = (n // self.batch_size) * self.batch_size if self.drop_last else n
         return [idx[i : min(i + self.batch_size, stop)] for i in range(0, stop, self.batch_size)]
 
+    def _fetch(self, b: torch.Tensor, slot: Optional[dict] = None):
+        """One batch fetch; CSR variables yield (values, offsets) tuples."""
+        store = self.store
+        if self.is_csr:
+            cap = b.numel() * self._max_len
+            buf = None if slot is None else slot.get("data")
+            if buf is None or buf.shape[0] < cap:
+                buf = torch.empty(
+                    (cap, store._meta(self.name)["disp"]),
+                    dtype=store._meta(self.name)["dtype"], device=store.device,
+                )
+            data = store.get_csr(self.name, b, out=buf)  # (values, offsets)
+        else:
+            buf = None if slot is None else slot.get("data")
+            data = store.get_batch(self.name, b, out=buf, dtype=self.out_dtype)
+        if slot is not None:
+            slot["data"] = data[0] if self.is_csr else data
+        if self.label_name is None:
+            return data, None
+        lbuf = None if slot is None else slot.get("label")
+        label = store.get_batch(self.label_name, b, out=lbuf, dtype=self.label_dtype)
+        if slot is not None:
+            slot["label"] = label
+        return data, label
+
     def __iter__(self) -> Iterator:
         if self.store.mode != "hip":
             for b in self._batches(self.indices):
-                data = self.store.get_batch(self.name, b, dtype=self.out_dtype)
-                if self.label_name is None:
-                    yield data
-                else:
-                    yield data, self.store.get_batch(self.label_name, b, dtype=self.label_dtype)
+                data, label = self._fetch(b)
+                yield data if label is None else (data, label)
             return
         yield from self._iter_hip()
 
@@ -105,14 +134,8 @@ class PrefetchLoader:
             with torch.cuda.stream(side):
                 if slot["free_recorded"]:
                     side.wait_event(slot["free"])
-                slot["data"] = store.get_batch(
-                    self.name, batches[j], out=slot["data"], dtype=self.out_dtype
-                )
-                if self.label_name is not None:
-                    slot["label"] = store.get_batch(
-                        self.label_name, batches[j], out=slot["label"],
-                        dtype=self.label_dtype,
-                    )
+                data, label = self._fetch(batches[j], slot)
+                slot["yield"] = data if label is None else (data, label)
                 slot["ready"].record(side)
 
         for j in range(min(self.depth, nb)):
@@ -122,10 +145,7 @@ class PrefetchLoader:
             slot = slots[i % self.depth]
             cur = torch.cuda.current_stream(device)
             cur.wait_event(slot["ready"])
-            if self.label_name is None:
-                yield slot["data"]
-            else:
-                yield slot["data"], slot["label"]
+            yield slot["yield"]
             # the consumer has enqueued its use of the buffers on the current
             # stream by the time it asks for the next batch
             cur = torch.cuda.current_stream(device)

@@ -334,3 +334,29 @@ def test_reshuffle_odd_rows_gpu(store, dtype, dim):
     out = store.get_batch(f"ro{dim}{str(dtype)[-4:]}", list(range(n)))
     torch.cuda.synchronize()
     assert torch.equal(out.cpu().view(torch.uint8), arr[perm].view(torch.uint8))
+
+
+def test_prefetch_csr_gpu(store):
+    from ddstore_amd import PrefetchLoader
+
+    rng = np.random.default_rng(4)
+    lengths = rng.integers(1, 12, size=300)
+    vals = torch.cat(
+        [torch.full((int(l), 2), float(i)) for i, l in enumerate(lengths)]
+    )
+    store.add_csr("pcg", vals, lengths)
+    store.add("pcl", torch.arange(300, dtype=torch.int64).unsqueeze(1))
+    order = rng.permutation(300)
+    seen = 0
+    for (v, off), y in PrefetchLoader(store, "pcg", order, batch_size=64,
+                                      label_name="pcl", depth=3):
+        torch.cuda.synchronize()
+        off_h = off.cpu().tolist()
+        y_h = y.view(-1).cpu().tolist()
+        for k in range(len(off_h) - 1):
+            g = int(order[seen + k])
+            seg = v.cpu()[off_h[k] : off_h[k + 1]]
+            assert y_h[k] == g
+            assert seg.shape[0] == lengths[g] and (seg == float(g)).all()
+        seen += len(off_h) - 1
+    assert seen == 300

@@ -299,3 +299,21 @@ def test_affine_gather_cpu(store):
     assert torch.allclose(out, ref)
     with pytest.raises(TypeError, match="affine output"):
         store.get_batch("af", [0], dtype=torch.int32, affine=(1.0, 0.0))
+
+
+def test_prefetch_csr_cpu(store):
+    from ddstore_amd import PrefetchLoader
+
+    lengths = [3, 1, 4, 2, 5, 2, 1, 6]
+    vals = np.concatenate(
+        [np.full(l, i, dtype=np.float32) for i, l in enumerate(lengths)]
+    ).reshape(-1, 1)
+    store.add_csr("pc", vals, lengths)
+    order = [7, 0, 3, 5, 1, 2, 6, 4]
+    got = []
+    for (v, off) in PrefetchLoader(store, "pc", order, batch_size=3):
+        off = off.tolist()
+        for k in range(len(off) - 1):
+            got.append(v[off[k] : off[k + 1], 0].tolist())
+    for k, g in enumerate(order):
+        assert got[k] == [float(g)] * lengths[g], (k, g)
