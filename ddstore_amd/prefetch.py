@@ -92,12 +92,16 @@ class PrefetchLoader:
             data = store.get_csr(self.name, b, out=buf)  # (values, offsets)
         else:
             buf = None if slot is None else slot.get("data")
+            if buf is not None and buf.shape[0] != b.numel():
+                buf = None  # ragged final batch: allocate a matching buffer
             data = store.get_batch(self.name, b, out=buf, dtype=self.out_dtype)
         if slot is not None:
             slot["data"] = data[0] if self.is_csr else data
         if self.label_name is None:
             return data, None
         lbuf = None if slot is None else slot.get("label")
+        if lbuf is not None and lbuf.shape[0] != b.numel():
+            lbuf = None
         label = store.get_batch(self.label_name, b, out=lbuf, dtype=self.label_dtype)
         if slot is not None:
             slot["label"] = label

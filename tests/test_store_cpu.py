@@ -317,3 +317,13 @@ def test_prefetch_csr_cpu(store):
             got.append(v[off[k] : off[k + 1], 0].tolist())
     for k, g in enumerate(order):
         assert got[k] == [float(g)] * lengths[g], (k, g)
+
+
+def test_prefetch_ragged_tail(store):
+    # final batch smaller than batch_size must not reuse a full-size buffer
+    from ddstore_amd import PrefetchLoader
+
+    arr = np.arange(50, dtype=np.float32).reshape(50, 1)
+    store.add("rt", arr)
+    sizes = [b.shape[0] for b in PrefetchLoader(store, "rt", np.arange(50), 16)]
+    assert sizes == [16, 16, 16, 2]
