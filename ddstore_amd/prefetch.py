@@ -48,7 +48,9 @@ class PrefetchLoader:
         label_dtype: Optional[torch.dtype] = None,
         depth: Optional[int] = None,
         drop_last: bool = False,
+        affine: Optional[tuple] = None,
     ):
+        self.affine = affine
         if depth is None:
             depth = int(os.environ.get("DDSTORE_PREFETCH_DEPTH", "2"))
         self.store = store
@@ -94,7 +96,12 @@ class PrefetchLoader:
             buf = None if slot is None else slot.get("data")
             if buf is not None and buf.shape[0] != b.numel():
                 buf = None  # ragged final batch: allocate a matching buffer
-            data = store.get_batch(self.name, b, out=buf, dtype=self.out_dtype)
+            if self.affine is not None and buf is None:
+                buf = torch.empty((b.numel(), store._meta(self.name)["disp"]),
+                                  dtype=self.out_dtype or torch.float32,
+                                  device=store.device)
+            data = store.get_batch(self.name, b, out=buf, dtype=self.out_dtype,
+                                   affine=self.affine)
         if slot is not None:
             slot["data"] = data[0] if self.is_csr else data
         if self.label_name is None:

@@ -24,7 +24,7 @@ import torch.nn as nn
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
-from ddstore_amd import DDStore  # noqa: E402
+from ddstore_amd import DDStore, PrefetchLoader  # noqa: E402
 
 FEAT = 16
 
@@ -90,11 +90,13 @@ def main():
         mine = torch.randperm(ntotal, generator=gs)[rank::world]
         correct, seen, tot_loss, nb = 0, 0, 0.0, 0
         store.epoch_begin()
-        for i in range(0, mine.numel() - args.batch_size + 1, args.batch_size):
-            bidx = mine[i : i + args.batch_size]
-            values, offsets = store.get_csr("graphs", bidx)     # packed nodes
-            y = store.get_batch("labels", bidx).view(-1).long()  # aligned labels
-            x = segment_mean(values, offsets).to(torch.bfloat16)
+        # CSR prefetch: next batch's variable-length gather runs on a side
+        # HIP stream while this batch trains
+        loader = PrefetchLoader(store, "graphs", mine, args.batch_size,
+                                label_name="labels", drop_last=True)
+        for (values, offsets), y in loader:
+            y = y.view(-1).long()
+            x = segment_mean(values[: int(offsets[-1])], offsets).to(torch.bfloat16)
             opt.zero_grad(set_to_none=True)
             logits = model(x)
             loss = lossf(logits.float(), y)
