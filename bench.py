@@ -42,7 +42,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--rows", type=int, default=2 * 1024 * 1024, help="rows per rank")
     p.add_argument("--dim", type=int, default=128, help="f32 elems per row (128 -> 512 B rows)")
-    p.add_argument("--batch", type=int, default=131072, help="rows fetched per rank per step")
+    p.add_argument("--batch", type=int, default=262144, help="rows fetched per rank per step")
     p.add_argument("--hidden", type=int, default=1024, help="MLP hidden width")
     p.add_argument("--mode", choices=["fetch", "train", "csr"], default="fetch",
                    help="fetch = global-shuffle sample-fetch throughput, the "
