@@ -30,7 +30,6 @@ import json
 import os
 import time
 
-import numpy as np
 import torch
 import torch.distributed as dist
 
