@@ -416,6 +416,40 @@ public:
         v.bytes_gathered += total_elems * v.row_elems * v.itemsize;
     }
 
+    // One-call CSR fetch: lens kernel -> cumsum -> gather, no host round
+    // trips between stages. `out` is a capacity buffer (>= worst-case batch
+    // elements); returns the [n+1] element-offset tensor (device).
+    at::Tensor gather_csr_fast(const std::string& name, const at::Tensor& idx,
+                               at::Tensor out) {
+        RoctxRange rr_("ddstore::gather_csr_fast");
+        DeviceVar& v = var(name);
+        TORCH_CHECK(v.is_csr, "ddstore gather_csr: not a CSR variable");
+        check_peers(v);
+        check_idx(idx);
+        const int64_t nidx = idx.numel();
+        TORCH_CHECK(out.is_contiguous() && out.device().is_cuda() &&
+                        dds_type_of(out) == v.dds_t,
+                    "ddstore gather_csr: bad output tensor");
+        auto opts = at::TensorOptions().dtype(at::kLong).device(idx.device());
+        at::Tensor off = at::zeros({nidx + 1}, opts);
+        at::Tensor lens = at::empty({nidx}, opts);
+        ddstore::csr_lens(stream(), v.d_goff, idx.data_ptr<int64_t>(), nidx,
+                          v.prefix[nparts_], lens.data_ptr<int64_t>(), v.d_oob);
+        at::Tensor off_tail = off.slice(0, 1, nidx + 1);
+        at::cumsum_out(off_tail, lens, 0);
+        const int64_t cap = out.numel() / std::max<int64_t>(v.row_elems, 1);
+        ddstore::gather_csr(stream(), (const void* const*)v.d_peers, v.d_prefix,
+                            v.d_elem_prefix, nparts_, v.d_goff,
+                            idx.data_ptr<int64_t>(), nidx,
+                            off.data_ptr<int64_t>(),
+                            v.row_elems * v.itemsize, cap,
+                            out.data_ptr(), v.d_oob);
+        v.n_gather += 1;
+        v.rows_gathered += nidx;
+        v.bytes_gathered += cap * v.row_elems * v.itemsize;
+        return off;
+    }
+
     void csr_lens(const std::string& name, const at::Tensor& idx, at::Tensor lens) {
         DeviceVar& v = var(name);
         TORCH_CHECK(v.is_csr, "ddstore csr_lens: not a CSR variable");
@@ -941,6 +975,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("gather_affine", &DeviceStore::gather_affine)
         .def("gather_csr", &DeviceStore::gather_csr)
         .def("csr_lens", &DeviceStore::csr_lens)
+        .def("gather_csr_fast", &DeviceStore::gather_csr_fast)
         .def("scatter_local", &DeviceStore::scatter_local)
         .def("local_shard", &DeviceStore::local_shard)
         .def("epoch_begin", &DeviceStore::epoch_begin)

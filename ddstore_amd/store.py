@@ -343,6 +343,12 @@ class DDStore:
             raise ValueError(f"ddstore get_csr: '{name}' is not a CSR variable")
         idx = torch.as_tensor(indices, dtype=torch.int64)
         idx = idx.to(self.device, non_blocking=True).contiguous()
+        if self.mode == "hip" and out is not None:
+            # capacity-buffer hot path: ONE native call (lens kernel ->
+            # device cumsum -> gather), no host sync, no Python between stages
+            self._account(meta, idx)
+            out_off = self._backend.gather_csr_fast(name, idx, out)
+            return out, out_off
         out_off = torch.zeros(idx.numel() + 1, dtype=torch.int64, device=self.device)
         if self.mode == "hip":
             lens = torch.empty(idx.numel(), dtype=torch.int64, device=self.device)
@@ -355,8 +361,6 @@ class DDStore:
             total = int(out_off[-1].item())
             out = torch.empty((total, meta["disp"]), dtype=meta["dtype"], device=self.device)
         else:
-            # caller-provided capacity buffer: no device sync in the hot loop
-            # (the actual packed length is out_off[-1], on device)
             total = out.numel() // meta["disp"]
         self._account(meta, idx)
         self._backend.gather_csr(name, idx, out_off, out, total)
