@@ -24,938 +24,10 @@
 // is issued by the Python layer (torch.distributed), while this class does
 // the local fence (hipStreamSynchronize of the store's stream).
 
-#include <torch/extension.h>
+#include "ddstore.hpp"
 
-#include <hip/hip_runtime.h>
-#include <c10/hip/HIPStream.h>
-#include <ATen/Parallel.h>
-
-#include <fcntl.h>
-#include <sys/mman.h>
-#include <unistd.h>
-
-#include <cstring>
-#include <map>
-#include <string>
-#include <vector>
-
-#include "ddstore_kernels.h"
-
-// roctx ranges so rocprofv3 --marker-trace attributes store phases
-// (the reference has no tracing at all)
-#include <roctracer/roctx.h>
-
-namespace {
-
-struct RoctxRange {
-    explicit RoctxRange(const char* name) { roctxRangePushA(name); }
-    ~RoctxRange() { roctxRangePop(); }
-};
-
-#define HIP_CHECK(expr)                                                        \
-    do {                                                                       \
-        hipError_t _e = (expr);                                                \
-        TORCH_CHECK(_e == hipSuccess, "ddstore HIP error: ",                   \
-                    hipGetErrorString(_e), " at " #expr);                      \
-    } while (0)
-
-int dds_type_of(const at::Tensor& t) {
-    switch (t.scalar_type()) {
-        case at::kByte: return DDS_U8;
-        case at::kBool: return DDS_U8;  // stored/moved as u8 (0/1), NumPy layout
-        case at::kInt: return DDS_I32;
-        case at::kLong: return DDS_I64;
-        case at::kFloat: return DDS_F32;
-        case at::kDouble: return DDS_F64;
-        case at::kHalf: return DDS_F16;
-        case at::kBFloat16: return DDS_BF16;
-        case at::kFloat8_e4m3fn: return DDS_F8E4M3;  // OCP fn, matches gfx950
-        case at::kFloat8_e5m2: return DDS_F8E5M2;
-        default:
-            TORCH_CHECK(false, "ddstore: unsupported dtype ", t.scalar_type());
-    }
-}
-
-int64_t dds_itemsize(int t) {
-    switch (t) {
-        case DDS_U8: case DDS_F8E4M3: case DDS_F8E5M2: return 1;
-        case DDS_F16: case DDS_BF16: return 2;
-        case DDS_I32: case DDS_F32: return 4;
-        default: return 8;
-    }
-}
-
-std::vector<int64_t> make_prefix(const std::vector<int64_t>& counts) {
-    std::vector<int64_t> prefix(counts.size() + 1, 0);
-    for (size_t i = 0; i < counts.size(); ++i) prefix[i + 1] = prefix[i] + counts[i];
-    return prefix;
-}
-
-// Host-side owner lookup over the prefix directory (binary search; the
-// reference's linear `sortedsearch` is src/ddstore.cxx:5-17).
-int owner_of_host(const std::vector<int64_t>& prefix, int64_t row) {
-    int lo = 0, hi = (int)prefix.size() - 2;
-    while (lo < hi) {
-        int mid = (lo + hi + 1) >> 1;
-        if (prefix[mid] <= row) lo = mid; else hi = mid - 1;
-    }
-    return lo;
-}
-
-struct EpochFSM {
-    bool fence_active = false;
-    void begin() {
-        TORCH_CHECK(!fence_active, "ddstore: epoch already began");  // ref ddstore.cxx:57
-        fence_active = true;
-    }
-    void end() {
-        TORCH_CHECK(fence_active, "ddstore: epoch has not begun");   // ref ddstore.cxx:71
-        fence_active = false;
-    }
-};
-
-// ===========================================================================
-// DeviceStore
-// ===========================================================================
-
-struct DeviceVar {
-    bool active = false;
-    bool is_csr = false;
-    int dds_t = 0;
-    at::ScalarType st = at::kFloat;
-    int64_t row_elems = 0;    // "disp" in reference terms
-    int64_t itemsize = 0;
-    int64_t nrows_local = 0;  // rows for fixed-stride; samples for CSR
-    std::vector<int64_t> prefix;
-    void* base = nullptr;
-    size_t base_bytes = 0;
-    std::vector<void*> peers;
-    std::vector<char> opened;    // 1 where hipIpcOpenMemHandle was used
-    void** d_peers = nullptr;
-    int64_t* d_prefix = nullptr;
-    unsigned long long* d_oob = nullptr;  // in-kernel OOB-index counter
-    // CSR extras
-    std::vector<int64_t> elem_prefix;
-    int64_t* d_elem_prefix = nullptr;
-    int64_t* d_goff = nullptr;   // replicated global element offsets [ntotal+1]
-    int64_t nelems_local = 0;
-    // stats
-    int64_t n_gather = 0, rows_gathered = 0, bytes_gathered = 0;
-};
-
-class DeviceStore {
-public:
-    DeviceStore(int device, int rank, int nparts)
-        : device_(device), rank_(rank), nparts_(nparts) {
-        TORCH_CHECK(nparts >= 1 && nparts <= DDS_MAX_PARTS,
-                    "ddstore: world size must be in [1, ", DDS_MAX_PARTS, "]");
-        TORCH_CHECK(rank >= 0 && rank < nparts, "ddstore: bad rank");
-    }
-    ~DeviceStore() { free_all_noexcept(); }
-
-    hipStream_t stream() const {
-        return c10::hip::getCurrentHIPStream(device_).stream();
-    }
-
-    void add(const std::string& name, const at::Tensor& src, int64_t nrows,
-             int64_t row_elems, std::vector<int64_t> nrows_all) {
-        check_new(name);
-        TORCH_CHECK(src.is_contiguous(), "ddstore add: array must be C-contiguous");
-        TORCH_CHECK(src.numel() == nrows * row_elems, "ddstore add: shape mismatch");
-        DeviceVar v;
-        v.st = src.scalar_type();
-        v.dds_t = dds_type_of(src);
-        v.itemsize = dds_itemsize(v.dds_t);
-        v.row_elems = row_elems;
-        v.nrows_local = nrows;
-        v.prefix = make_prefix(nrows_all);
-        alloc_base(v, (size_t)(nrows * row_elems * v.itemsize));
-        try {
-            ingest(v.base, src, 0);
-        } catch (...) {
-            release(v);
-            throw;
-        }
-        v.active = true;
-        vars_[name] = std::move(v);
-    }
-
-    void init(const std::string& name, int64_t nrows, int64_t row_elems,
-              at::ScalarType st, std::vector<int64_t> nrows_all) {
-        // reference: pre-allocate zeroed shard, fill later via update
-        // (ddstore.hpp:110-179; README.md:107 -- no epoch required)
-        check_new(name);
-        at::Tensor proto = at::empty({0}, at::TensorOptions().dtype(st));
-        DeviceVar v;
-        v.st = st;
-        v.dds_t = dds_type_of(proto);
-        v.itemsize = dds_itemsize(v.dds_t);
-        v.row_elems = row_elems;
-        v.nrows_local = nrows;
-        v.prefix = make_prefix(nrows_all);
-        alloc_base(v, (size_t)(nrows * row_elems * v.itemsize));
-        try {
-            HIP_CHECK(hipMemsetAsync(v.base, 0, v.base_bytes, stream()));
-            HIP_CHECK(hipStreamSynchronize(stream()));
-        } catch (...) {
-            release(v);
-            throw;
-        }
-        v.active = true;
-        vars_[name] = std::move(v);
-    }
-
-    void add_csr(const std::string& name, const at::Tensor& values,
-                 int64_t nsamples, int64_t nelems, int64_t row_elems,
-                 std::vector<int64_t> nsamples_all, std::vector<int64_t> nelems_all,
-                 const at::Tensor& goff_cpu) {
-        // CSR first-class layout: the reference layers variable-length records
-        // on an element-addressed store with disp=1 (HydraGNN
-        // pattern); here offsets are replicated per GPU and the gather kernel
-        // packs whole samples.
-        check_new(name);
-        TORCH_CHECK(values.is_contiguous(), "ddstore add_csr: values must be contiguous");
-        TORCH_CHECK(values.numel() == nelems * row_elems, "ddstore add_csr: shape mismatch");
-        DeviceVar v;
-        v.is_csr = true;
-        v.st = values.scalar_type();
-        v.dds_t = dds_type_of(values);
-        v.itemsize = dds_itemsize(v.dds_t);
-        v.row_elems = row_elems;     // feature width per element
-        v.nrows_local = nsamples;
-        v.nelems_local = nelems;
-        v.prefix = make_prefix(nsamples_all);
-        v.elem_prefix = make_prefix(nelems_all);
-        TORCH_CHECK(goff_cpu.scalar_type() == at::kLong && goff_cpu.is_contiguous() &&
-                        goff_cpu.device().is_cpu() &&
-                        goff_cpu.numel() == v.prefix[nparts_] + 1,
-                    "ddstore add_csr: bad global offsets");
-        alloc_base(v, (size_t)(nelems * row_elems * v.itemsize));
-        try {
-            ingest(v.base, values, 0);
-            size_t gbytes = (size_t)goff_cpu.numel() * 8;
-            HIP_CHECK(hipMalloc((void**)&v.d_goff, gbytes));
-            HIP_CHECK(hipMemcpy(v.d_goff, goff_cpu.data_ptr<int64_t>(), gbytes,
-                                hipMemcpyHostToDevice));
-        } catch (...) {
-            release(v);
-            throw;
-        }
-        v.active = true;
-        vars_[name] = std::move(v);
-    }
-
-    py::bytes ipc_handle(const std::string& name) {
-        DeviceVar& v = var(name);
-        hipIpcMemHandle_t h;
-        HIP_CHECK(hipIpcGetMemHandle(&h, v.base));
-        return py::bytes(reinterpret_cast<const char*>(&h), sizeof(h));
-    }
-
-    // Finalize a variable: map peer shards (empty bytes => use own base) and
-    // mirror the directory + pointer table to device memory.
-    // (reference analog: the MPI_Win_create collective, ddstore.hpp:56-62, or
-    // the libfabric handshake's 3x Allgather, common.cxx:285-302)
-    void open_peers(const std::string& name, const std::vector<std::string>& handles) {
-        DeviceVar& v = var(name);
-        TORCH_CHECK((int)handles.size() == nparts_, "ddstore open_peers: size mismatch");
-        v.peers.assign(nparts_, nullptr);
-        v.opened.assign(nparts_, 0);
-        for (int r = 0; r < nparts_; ++r) {
-            if (r == rank_) {
-                v.peers[r] = v.base;
-            } else {
-                TORCH_CHECK(handles[r].size() == sizeof(hipIpcMemHandle_t),
-                            "ddstore open_peers: bad handle bytes for rank ", r);
-                hipIpcMemHandle_t h;
-                std::memcpy(&h, handles[r].data(), sizeof(h));
-                void* p = nullptr;
-                HIP_CHECK(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess));
-                v.peers[r] = p;
-                v.opened[r] = 1;
-            }
-        }
-        // one device block: [peer ptrs | prefix | elem_prefix? | oob counter]
-        size_t nb = nparts_ * sizeof(void*) + (nparts_ + 1) * 8 +
-                    (v.is_csr ? (nparts_ + 1) * 8 : 0) + 8;
-        HIP_CHECK(hipMalloc((void**)&v.d_peers, nb));
-        v.d_prefix = reinterpret_cast<int64_t*>(v.d_peers + nparts_);
-        HIP_CHECK(hipMemcpy(v.d_peers, v.peers.data(), nparts_ * sizeof(void*),
-                            hipMemcpyHostToDevice));
-        HIP_CHECK(hipMemcpy(v.d_prefix, v.prefix.data(), (nparts_ + 1) * 8,
-                            hipMemcpyHostToDevice));
-        int64_t* tail = v.d_prefix + nparts_ + 1;
-        if (v.is_csr) {
-            v.d_elem_prefix = tail;
-            HIP_CHECK(hipMemcpy(v.d_elem_prefix, v.elem_prefix.data(),
-                                (nparts_ + 1) * 8, hipMemcpyHostToDevice));
-            tail += nparts_ + 1;
-        }
-        v.d_oob = reinterpret_cast<unsigned long long*>(tail);
-        HIP_CHECK(hipMemset(v.d_oob, 0, 8));
-    }
-
-    void update(const std::string& name, const at::Tensor& src, int64_t offset) {
-        // purely local fill at row offset (reference ddstore.hpp:181-195)
-        DeviceVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore update: not supported for CSR variables");
-        TORCH_CHECK(src.is_contiguous(), "ddstore update: array must be C-contiguous");
-        TORCH_CHECK(dds_itemsize(dds_type_of(src)) == v.itemsize,
-                    "ddstore update: itemsize mismatch");
-        int64_t nrows = src.numel() / v.row_elems;
-        TORCH_CHECK(src.numel() == nrows * v.row_elems, "ddstore update: shape mismatch");
-        TORCH_CHECK(offset >= 0 && offset + nrows <= v.nrows_local,
-                    "ddstore update: out of range");
-        ingest(v.base, src, offset * v.row_elems * v.itemsize);
-    }
-
-    void get_range(const std::string& name, int64_t start, int64_t count,
-                   at::Tensor out) {
-        // reference get semantics: dense [start, start+count) rows from ONE
-        // owner; throws if the range crosses a shard boundary
-        // (ddstore.hpp:197-248)
-        DeviceVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore get: use get_csr for CSR variables");
-        TORCH_CHECK(out.is_contiguous(), "ddstore get: output must be C-contiguous");
-        TORCH_CHECK(dds_itemsize(dds_type_of(out)) == v.itemsize,
-                    "ddstore get: itemsize mismatch");
-        TORCH_CHECK(out.numel() == count * v.row_elems, "ddstore get: shape mismatch");
-        check_peers(v);
-        int64_t ntotal = v.prefix[nparts_];
-        TORCH_CHECK(start >= 0 && start < ntotal, "Invalid start on target");
-        int target = owner_of_host(v.prefix, start);
-        TORCH_CHECK(start + count <= v.prefix[target + 1], "Invalid count on target");
-        const char* src = (const char*)v.peers[target] +
-                          (start - v.prefix[target]) * v.row_elems * v.itemsize;
-        size_t nb = (size_t)count * v.row_elems * v.itemsize;
-        if (out.device().is_cpu()) {
-            HIP_CHECK(hipStreamSynchronize(stream()));
-            HIP_CHECK(hipMemcpy(out.data_ptr(), src, nb, hipMemcpyDeviceToHost));
-        } else {
-            TORCH_CHECK(out.device().index() == device_, "ddstore get: wrong device");
-            HIP_CHECK(hipMemcpyAsync(out.data_ptr(), src, nb,
-                                     hipMemcpyDeviceToDevice, stream()));
-        }
-        v.n_gather += 1;
-        v.rows_gathered += count;
-        v.bytes_gathered += nb;
-    }
-
-    void gather(const std::string& name, const at::Tensor& idx, at::Tensor out) {
-        // the batched hot path: one launch per minibatch
-        RoctxRange rr_("ddstore::gather");
-        DeviceVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore gather: use gather_csr for CSR variables");
-        check_peers(v);
-        check_idx(idx);
-        int64_t nidx = idx.numel();
-        TORCH_CHECK(out.is_contiguous() && out.device().is_cuda() &&
-                        out.device().index() == device_,
-                    "ddstore gather: output must be a contiguous tensor on the store device");
-        TORCH_CHECK(out.numel() == nidx * v.row_elems, "ddstore gather: shape mismatch");
-        // byte move only for the SAME dds type (bool aliases u8); any other
-        // dtype pair converts numerically -- a same-size pair like f16->bf16
-        // must NOT silently reinterpret bits (the CPU path converts too)
-        int out_t = dds_type_of(out);
-        ddstore::gather_rows(stream(), (const void* const*)v.d_peers, v.d_prefix,
-                             nparts_, idx.data_ptr<int64_t>(), nidx, v.row_elems,
-                             v.dds_t, out_t, out.data_ptr(), v.d_oob);
-        v.n_gather += 1;
-        v.rows_gathered += nidx;
-        v.bytes_gathered += nidx * v.row_elems * v.itemsize;
-    }
-
-    void gather_affine(const std::string& name, const at::Tensor& idx,
-                       at::Tensor out, double scale, double shift) {
-        RoctxRange rr_("ddstore::gather_affine");
-        DeviceVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore gather: use gather_csr for CSR variables");
-        check_peers(v);
-        check_idx(idx);
-        int64_t nidx = idx.numel();
-        TORCH_CHECK(out.is_contiguous() && out.device().is_cuda() &&
-                        out.device().index() == device_,
-                    "ddstore gather: output must be a contiguous tensor on the store device");
-        TORCH_CHECK(out.numel() == nidx * v.row_elems, "ddstore gather: shape mismatch");
-        int out_t = dds_type_of(out);
-        TORCH_CHECK(out_t == DDS_F32 || out_t == DDS_F16 || out_t == DDS_BF16,
-                    "ddstore gather: affine output must be f32/f16/bf16");
-        ddstore::gather_rows_affine(stream(), (const void* const*)v.d_peers,
-                                    v.d_prefix, nparts_, idx.data_ptr<int64_t>(),
-                                    nidx, v.row_elems, v.dds_t, out_t,
-                                    (float)scale, (float)shift, out.data_ptr(),
-                                    v.d_oob);
-        v.n_gather += 1;
-        v.rows_gathered += nidx;
-        v.bytes_gathered += nidx * v.row_elems * v.itemsize;
-    }
-
-    void gather_csr(const std::string& name, const at::Tensor& idx,
-                    const at::Tensor& out_off, at::Tensor out, int64_t total_elems) {
-        RoctxRange rr_("ddstore::gather_csr");
-        DeviceVar& v = var(name);
-        TORCH_CHECK(v.is_csr, "ddstore gather_csr: not a CSR variable");
-        check_peers(v);
-        check_idx(idx);
-        check_idx(out_off);
-        int64_t nidx = idx.numel();
-        TORCH_CHECK(out_off.numel() == nidx + 1, "ddstore gather_csr: bad out_off");
-        TORCH_CHECK(out.is_contiguous() && out.device().is_cuda() &&
-                        dds_type_of(out) == v.dds_t,
-                    "ddstore gather_csr: bad output tensor");
-        TORCH_CHECK(out.numel() >= total_elems * v.row_elems,
-                    "ddstore gather_csr: output too small");
-        ddstore::gather_csr(stream(), (const void* const*)v.d_peers, v.d_prefix,
-                            v.d_elem_prefix, nparts_, v.d_goff,
-                            idx.data_ptr<int64_t>(), nidx,
-                            out_off.data_ptr<int64_t>(),
-                            v.row_elems * v.itemsize, total_elems,
-                            out.data_ptr(), v.d_oob);
-        v.n_gather += 1;
-        v.rows_gathered += nidx;
-        v.bytes_gathered += total_elems * v.row_elems * v.itemsize;
-    }
-
-    // One-call CSR fetch: lens kernel -> cumsum -> gather, no host round
-    // trips between stages. `out` is a capacity buffer (>= worst-case batch
-    // elements); returns the [n+1] element-offset tensor (device).
-    at::Tensor gather_csr_fast(const std::string& name, const at::Tensor& idx,
-                               at::Tensor out) {
-        RoctxRange rr_("ddstore::gather_csr_fast");
-        DeviceVar& v = var(name);
-        TORCH_CHECK(v.is_csr, "ddstore gather_csr: not a CSR variable");
-        check_peers(v);
-        check_idx(idx);
-        const int64_t nidx = idx.numel();
-        TORCH_CHECK(out.is_contiguous() && out.device().is_cuda() &&
-                        dds_type_of(out) == v.dds_t,
-                    "ddstore gather_csr: bad output tensor");
-        auto opts = at::TensorOptions().dtype(at::kLong).device(idx.device());
-        at::Tensor off = at::zeros({nidx + 1}, opts);
-        at::Tensor lens = at::empty({nidx}, opts);
-        ddstore::csr_lens(stream(), v.d_goff, idx.data_ptr<int64_t>(), nidx,
-                          v.prefix[nparts_], lens.data_ptr<int64_t>(), v.d_oob);
-        at::Tensor off_tail = off.slice(0, 1, nidx + 1);
-        at::cumsum_out(off_tail, lens, 0);
-        const int64_t cap = out.numel() / std::max<int64_t>(v.row_elems, 1);
-        ddstore::gather_csr(stream(), (const void* const*)v.d_peers, v.d_prefix,
-                            v.d_elem_prefix, nparts_, v.d_goff,
-                            idx.data_ptr<int64_t>(), nidx,
-                            off.data_ptr<int64_t>(),
-                            v.row_elems * v.itemsize, cap,
-                            out.data_ptr(), v.d_oob);
-        v.n_gather += 1;
-        v.rows_gathered += nidx;
-        v.bytes_gathered += cap * v.row_elems * v.itemsize;
-        return off;
-    }
-
-    void csr_lens(const std::string& name, const at::Tensor& idx, at::Tensor lens) {
-        DeviceVar& v = var(name);
-        TORCH_CHECK(v.is_csr, "ddstore csr_lens: not a CSR variable");
-        check_peers(v);
-        check_idx(idx);
-        TORCH_CHECK(lens.scalar_type() == at::kLong && lens.is_contiguous() &&
-                        lens.device().is_cuda() && lens.numel() == idx.numel(),
-                    "ddstore csr_lens: bad lens tensor");
-        ddstore::csr_lens(stream(), v.d_goff, idx.data_ptr<int64_t>(), idx.numel(),
-                          v.prefix[nparts_], lens.data_ptr<int64_t>(), v.d_oob);
-    }
-
-    void scatter_local(const std::string& name, const at::Tensor& local_idx,
-                       const at::Tensor& src) {
-        // reshuffle placement: src row r -> local row local_idx[r]
-        DeviceVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore scatter_local: CSR not supported");
-        check_idx(local_idx);
-        TORCH_CHECK(src.is_contiguous() && src.device().is_cuda() &&
-                        dds_type_of(src) == v.dds_t,
-                    "ddstore scatter_local: bad src");
-        TORCH_CHECK(src.numel() == local_idx.numel() * v.row_elems,
-                    "ddstore scatter_local: shape mismatch");
-        check_peers(v);  // d_oob lives in the metadata block
-        ddstore::scatter_rows_local(stream(), v.base, v.nrows_local, v.row_elems,
-                                    v.dds_t, local_idx.data_ptr<int64_t>(),
-                                    local_idx.numel(), src.data_ptr(), v.d_oob);
-    }
-
-    // Zero-copy view of the local shard as a torch tensor (does NOT own).
-    at::Tensor local_shard(const std::string& name) {
-        DeviceVar& v = var(name);
-        int64_t n = v.is_csr ? v.nelems_local : v.nrows_local;
-        auto opts = at::TensorOptions().dtype(v.st).device(at::kCUDA, device_);
-        return at::from_blob(v.base, {n, v.row_elems}, opts);
-    }
-
-    void epoch_begin() {
-        fsm_.begin();
-        HIP_CHECK(hipStreamSynchronize(stream()));
-    }
-    void epoch_end() {
-        fsm_.end();
-        HIP_CHECK(hipStreamSynchronize(stream()));
-    }
-
-    py::dict query(const std::string& name) {
-        DeviceVar& v = var(name);
-        py::dict d;
-        d["nrows_local"] = v.nrows_local;
-        d["nrows_total"] = v.prefix[nparts_];
-        d["disp"] = v.row_elems;
-        d["itemsize"] = v.itemsize;
-        d["is_csr"] = v.is_csr;
-        d["prefix"] = v.prefix;
-        if (v.is_csr) {
-            d["nelems_local"] = v.nelems_local;
-            d["elem_prefix"] = v.elem_prefix;
-        }
-        d["n_gather"] = v.n_gather;
-        d["rows_gathered"] = v.rows_gathered;
-        d["bytes_gathered"] = v.bytes_gathered;
-        if (v.d_oob) {
-            unsigned long long oob = 0;
-            HIP_CHECK(hipStreamSynchronize(stream()));
-            HIP_CHECK(hipMemcpy(&oob, v.d_oob, 8, hipMemcpyDeviceToHost));
-            d["oob_skipped"] = (int64_t)oob;
-        }
-        return d;
-    }
-
-    bool has(const std::string& name) const { return vars_.count(name) > 0; }
-
-    void free_var(const std::string& name) {
-        auto it = vars_.find(name);
-        TORCH_CHECK(it != vars_.end(), "ddstore: unknown variable '", name, "'");
-        release(it->second);
-        vars_.erase(it);
-    }
-
-    void free_all() {
-        for (auto& kv : vars_) release(kv.second);
-        vars_.clear();
-    }
-
-private:
-    void free_all_noexcept() noexcept {
-        try { free_all(); } catch (...) { /* after runtime teardown */ }
-    }
-
-    void check_new(const std::string& name) {
-        TORCH_CHECK(!vars_.count(name), "ddstore: variable '", name, "' already exists");
-    }
-    DeviceVar& var(const std::string& name) {
-        auto it = vars_.find(name);
-        TORCH_CHECK(it != vars_.end(), "ddstore: unknown variable '", name, "'");
-        return it->second;
-    }
-    void check_peers(const DeviceVar& v) {
-        TORCH_CHECK(v.d_peers != nullptr,
-                    "ddstore: variable not finalized (open_peers not called)");
-    }
-    void check_idx(const at::Tensor& idx) {
-        TORCH_CHECK(idx.scalar_type() == at::kLong && idx.is_contiguous() &&
-                        idx.device().is_cuda() && idx.device().index() == device_,
-                    "ddstore: indices must be a contiguous int64 tensor on the store device");
-    }
-
-    void alloc_base(DeviceVar& v, size_t bytes) {
-        hipError_t e = hipSetDevice(device_);
-        TORCH_CHECK(e == hipSuccess, "ddstore: hipSetDevice failed");
-        v.base_bytes = bytes < 256 ? 256 : bytes;
-        HIP_CHECK(hipMalloc(&v.base, v.base_bytes));
-    }
-
-    void ingest(void* dst_base, const at::Tensor& src, int64_t byte_off) {
-        char* dst = (char*)dst_base + byte_off;
-        size_t nb = (size_t)src.numel() * src.element_size();
-        if (nb == 0) return;
-        if (src.device().is_cpu()) {
-            HIP_CHECK(hipMemcpy(dst, src.data_ptr(), nb, hipMemcpyHostToDevice));
-        } else {
-            TORCH_CHECK(src.device().index() == device_, "ddstore: wrong source device");
-            HIP_CHECK(hipMemcpyAsync(dst, src.data_ptr(), nb,
-                                     hipMemcpyDeviceToDevice, stream()));
-            HIP_CHECK(hipStreamSynchronize(stream()));
-        }
-    }
-
-    void release(DeviceVar& v) {
-        if (!v.active && !v.base) return;
-        for (int r = 0; r < (int)v.peers.size(); ++r)
-            if (v.opened.size() > (size_t)r && v.opened[r] && v.peers[r])
-                (void)hipIpcCloseMemHandle(v.peers[r]);
-        v.peers.clear();
-        if (v.d_peers) (void)hipFree(v.d_peers);
-        if (v.d_goff) (void)hipFree(v.d_goff);
-        if (v.base) (void)hipFree(v.base);
-        v.base = nullptr;
-        v.d_peers = nullptr;
-        v.d_goff = nullptr;
-        v.active = false;
-    }
-
-    int device_;
-    int rank_;
-    int nparts_;
-    EpochFSM fsm_;
-    std::map<std::string, DeviceVar> vars_;
-};
-
-// ===========================================================================
-// HostStore (POSIX shm, CPU compatibility path -- BASELINE config 1)
-// ===========================================================================
-
-struct HostVar {
-    bool active = false;
-    bool is_csr = false;
-    bool creator = false;
-    int dds_t = 0;
-    at::ScalarType st = at::kFloat;
-    int64_t row_elems = 0;
-    int64_t itemsize = 0;
-    int64_t nrows_local = 0;
-    int64_t nelems_local = 0;
-    std::vector<int64_t> prefix;
-    std::vector<int64_t> elem_prefix;
-    std::string shm_name;
-    void* base = nullptr;
-    size_t base_bytes = 0;
-    std::vector<void*> peers;
-    std::vector<size_t> peer_bytes;
-    std::vector<std::string> peer_names;
-    at::Tensor goff;  // CPU int64 [ntotal+1], replicated
-    int64_t n_gather = 0, rows_gathered = 0, bytes_gathered = 0;
-};
-
-class HostStore {
-public:
-    HostStore(const std::string& session, int rank, int nparts)
-        : session_(session), rank_(rank), nparts_(nparts) {
-        TORCH_CHECK(nparts >= 1 && nparts <= DDS_MAX_PARTS,
-                    "ddstore: world size must be in [1, ", DDS_MAX_PARTS, "]");
-    }
-    ~HostStore() {
-        try { free_all(); } catch (...) {}
-    }
-
-    std::string shm_name(const std::string& var) const {
-        return "/dds-" + session_ + "-" + var + "-" + std::to_string(rank_);
-    }
-
-    std::string add(const std::string& name, const at::Tensor& src, int64_t nrows,
-                    int64_t row_elems, std::vector<int64_t> nrows_all) {
-        check_new(name);
-        TORCH_CHECK(src.is_contiguous() && src.device().is_cpu(),
-                    "ddstore add: array must be C-contiguous on CPU");
-        TORCH_CHECK(src.numel() == nrows * row_elems, "ddstore add: shape mismatch");
-        HostVar v;
-        v.st = src.scalar_type();
-        v.dds_t = dds_type_of(src);
-        v.itemsize = dds_itemsize(v.dds_t);
-        v.row_elems = row_elems;
-        v.nrows_local = nrows;
-        v.prefix = make_prefix(nrows_all);
-        create_shm(v, name, (size_t)(nrows * row_elems * v.itemsize));
-        std::memcpy(v.base, src.data_ptr(), (size_t)src.numel() * v.itemsize);
-        v.active = true;
-        std::string n = v.shm_name;
-        vars_[name] = std::move(v);
-        return n;
-    }
-
-    std::string init(const std::string& name, int64_t nrows, int64_t row_elems,
-                     at::ScalarType st, std::vector<int64_t> nrows_all) {
-        check_new(name);
-        at::Tensor proto = at::empty({0}, at::TensorOptions().dtype(st));
-        HostVar v;
-        v.st = st;
-        v.dds_t = dds_type_of(proto);
-        v.itemsize = dds_itemsize(v.dds_t);
-        v.row_elems = row_elems;
-        v.nrows_local = nrows;
-        v.prefix = make_prefix(nrows_all);
-        create_shm(v, name, (size_t)(nrows * row_elems * v.itemsize));
-        std::memset(v.base, 0, v.base_bytes);
-        v.active = true;
-        std::string n = v.shm_name;
-        vars_[name] = std::move(v);
-        return n;
-    }
-
-    std::string add_csr(const std::string& name, const at::Tensor& values,
-                        int64_t nsamples, int64_t nelems, int64_t row_elems,
-                        std::vector<int64_t> nsamples_all,
-                        std::vector<int64_t> nelems_all, const at::Tensor& goff_cpu) {
-        check_new(name);
-        HostVar v;
-        v.is_csr = true;
-        v.st = values.scalar_type();
-        v.dds_t = dds_type_of(values);
-        v.itemsize = dds_itemsize(v.dds_t);
-        v.row_elems = row_elems;
-        v.nrows_local = nsamples;
-        v.nelems_local = nelems;
-        v.prefix = make_prefix(nsamples_all);
-        v.elem_prefix = make_prefix(nelems_all);
-        TORCH_CHECK(values.is_contiguous() && values.device().is_cpu(),
-                    "ddstore add_csr: values must be contiguous on CPU");
-        TORCH_CHECK(values.numel() == nelems * row_elems, "ddstore add_csr: shape mismatch");
-        TORCH_CHECK(goff_cpu.scalar_type() == at::kLong && goff_cpu.is_contiguous() &&
-                        goff_cpu.numel() == v.prefix[nparts_] + 1,
-                    "ddstore add_csr: bad global offsets");
-        v.goff = goff_cpu;
-        create_shm(v, name, (size_t)(nelems * row_elems * v.itemsize));
-        std::memcpy(v.base, values.data_ptr(), (size_t)values.numel() * v.itemsize);
-        v.active = true;
-        std::string n = v.shm_name;
-        vars_[name] = std::move(v);
-        return n;
-    }
-
-    void open_peers(const std::string& name, const std::vector<std::string>& names) {
-        HostVar& v = var(name);
-        TORCH_CHECK((int)names.size() == nparts_, "ddstore open_peers: size mismatch");
-        v.peers.assign(nparts_, nullptr);
-        v.peer_bytes.assign(nparts_, 0);
-        v.peer_names = names;
-        const std::vector<int64_t>& cnt_prefix = v.is_csr ? v.elem_prefix : v.prefix;
-        for (int r = 0; r < nparts_; ++r) {
-            size_t nb = (size_t)((cnt_prefix[r + 1] - cnt_prefix[r]) * v.row_elems *
-                                 v.itemsize);
-            if (nb < 1) nb = 1;
-            if (r == rank_) {
-                v.peers[r] = v.base;
-                v.peer_bytes[r] = 0;  // not mapped, do not munmap
-                continue;
-            }
-            int fd = shm_open(names[r].c_str(), O_RDONLY, 0600);
-            TORCH_CHECK(fd >= 0, "ddstore: shm_open failed for ", names[r]);
-            void* p = mmap(nullptr, nb, PROT_READ, MAP_SHARED, fd, 0);
-            close(fd);
-            TORCH_CHECK(p != MAP_FAILED, "ddstore: mmap failed for ", names[r]);
-            v.peers[r] = p;
-            v.peer_bytes[r] = nb;
-        }
-    }
-
-    void update(const std::string& name, const at::Tensor& src, int64_t offset) {
-        HostVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore update: not supported for CSR variables");
-        TORCH_CHECK(src.is_contiguous() && src.device().is_cpu(),
-                    "ddstore update: array must be C-contiguous on CPU");
-        TORCH_CHECK(dds_itemsize(dds_type_of(src)) == v.itemsize,
-                    "ddstore update: itemsize mismatch");
-        int64_t nrows = src.numel() / v.row_elems;
-        TORCH_CHECK(src.numel() == nrows * v.row_elems, "ddstore update: shape mismatch");
-        TORCH_CHECK(offset >= 0 && offset + nrows <= v.nrows_local,
-                    "ddstore update: out of range");
-        std::memcpy((char*)v.base + offset * v.row_elems * v.itemsize, src.data_ptr(),
-                    (size_t)src.numel() * v.itemsize);
-    }
-
-    void get_range(const std::string& name, int64_t start, int64_t count,
-                   at::Tensor out) {
-        HostVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore get: use get_csr for CSR variables");
-        TORCH_CHECK(out.is_contiguous() && out.device().is_cpu(),
-                    "ddstore get: output must be C-contiguous on CPU");
-        TORCH_CHECK(dds_itemsize(dds_type_of(out)) == v.itemsize,
-                    "ddstore get: itemsize mismatch");
-        TORCH_CHECK(out.numel() == count * v.row_elems, "ddstore get: shape mismatch");
-        check_peers(v);
-        int64_t ntotal = v.prefix[nparts_];
-        TORCH_CHECK(start >= 0 && start < ntotal, "Invalid start on target");
-        int target = owner_of_host(v.prefix, start);
-        TORCH_CHECK(start + count <= v.prefix[target + 1], "Invalid count on target");
-        const char* src = (const char*)v.peers[target] +
-                          (start - v.prefix[target]) * v.row_elems * v.itemsize;
-        size_t nb = (size_t)count * v.row_elems * v.itemsize;
-        std::memcpy(out.data_ptr(), src, nb);
-        v.n_gather += 1;
-        v.rows_gathered += count;
-        v.bytes_gathered += nb;
-    }
-
-    void gather(const std::string& name, const at::Tensor& idx, at::Tensor out) {
-        HostVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore gather: use gather_csr for CSR variables");
-        check_peers(v);
-        TORCH_CHECK(idx.scalar_type() == at::kLong && idx.is_contiguous() &&
-                        idx.device().is_cpu(),
-                    "ddstore gather: indices must be contiguous int64 on CPU");
-        int64_t nidx = idx.numel();
-        TORCH_CHECK(out.is_contiguous() && out.device().is_cpu() &&
-                        dds_itemsize(dds_type_of(out)) == v.itemsize,
-                    "ddstore gather: output must be contiguous CPU tensor of the store itemsize");
-        TORCH_CHECK(out.numel() == nidx * v.row_elems, "ddstore gather: shape mismatch");
-        const int64_t* ip = idx.data_ptr<int64_t>();
-        const int64_t rb = v.row_elems * v.itemsize;
-        char* op = (char*)out.data_ptr();
-        const int64_t ntotal = v.prefix[nparts_];
-        at::parallel_for(0, nidx, 1024, [&](int64_t b, int64_t e) {
-            for (int64_t i = b; i < e; ++i) {
-                int64_t g = ip[i];
-                TORCH_CHECK(g >= 0 && g < ntotal, "ddstore gather: index out of range");
-                int p = owner_of_host(v.prefix, g);
-                std::memcpy(op + i * rb,
-                            (const char*)v.peers[p] + (g - v.prefix[p]) * rb,
-                            (size_t)rb);
-            }
-        });
-        v.n_gather += 1;
-        v.rows_gathered += nidx;
-        v.bytes_gathered += nidx * rb;
-    }
-
-    void gather_csr(const std::string& name, const at::Tensor& idx,
-                    const at::Tensor& out_off, at::Tensor out, int64_t total_elems) {
-        HostVar& v = var(name);
-        TORCH_CHECK(v.is_csr, "ddstore gather_csr: not a CSR variable");
-        check_peers(v);
-        TORCH_CHECK(idx.scalar_type() == at::kLong && idx.is_contiguous() &&
-                        idx.device().is_cpu(), "ddstore gather_csr: bad indices");
-        TORCH_CHECK(out_off.scalar_type() == at::kLong && out_off.is_contiguous() &&
-                        out_off.numel() == idx.numel() + 1,
-                    "ddstore gather_csr: bad out_off");
-        TORCH_CHECK(out.is_contiguous() && out.device().is_cpu() &&
-                        dds_type_of(out) == v.dds_t &&
-                        out.numel() >= total_elems * v.row_elems,
-                    "ddstore gather_csr: bad output tensor");
-        const int64_t* ip = idx.data_ptr<int64_t>();
-        const int64_t* oo = out_off.data_ptr<int64_t>();
-        const int64_t* go = v.goff.data_ptr<int64_t>();
-        const int64_t eb = v.row_elems * v.itemsize;
-        char* op = (char*)out.data_ptr();
-        at::parallel_for(0, idx.numel(), 64, [&](int64_t b, int64_t e) {
-            for (int64_t s = b; s < e; ++s) {
-                int64_t g = ip[s];
-                int p = owner_of_host(v.prefix, g);
-                int64_t e0 = go[g], n = go[g + 1] - go[g];
-                std::memcpy(op + oo[s] * eb,
-                            (const char*)v.peers[p] + (e0 - v.elem_prefix[p]) * eb,
-                            (size_t)(n * eb));
-            }
-        });
-        v.n_gather += 1;
-        v.rows_gathered += idx.numel();
-        v.bytes_gathered += total_elems * eb;
-    }
-
-    void scatter_local(const std::string& name, const at::Tensor& local_idx,
-                       const at::Tensor& src) {
-        HostVar& v = var(name);
-        TORCH_CHECK(!v.is_csr, "ddstore scatter_local: CSR not supported");
-        TORCH_CHECK(local_idx.scalar_type() == at::kLong && local_idx.is_contiguous() &&
-                        local_idx.device().is_cpu(), "ddstore scatter_local: bad indices");
-        TORCH_CHECK(src.is_contiguous() && src.device().is_cpu() &&
-                        dds_type_of(src) == v.dds_t &&
-                        src.numel() == local_idx.numel() * v.row_elems,
-                    "ddstore scatter_local: bad src");
-        const int64_t* ip = local_idx.data_ptr<int64_t>();
-        const int64_t rb = v.row_elems * v.itemsize;
-        const char* sp = (const char*)src.data_ptr();
-        at::parallel_for(0, local_idx.numel(), 1024, [&](int64_t b, int64_t e) {
-            for (int64_t i = b; i < e; ++i) {
-                TORCH_CHECK(ip[i] >= 0 && ip[i] < v.nrows_local,
-                            "ddstore scatter_local: index out of range");
-                std::memcpy((char*)v.base + ip[i] * rb, sp + i * rb, (size_t)rb);
-            }
-        });
-    }
-
-    at::Tensor local_shard(const std::string& name) {
-        HostVar& v = var(name);
-        int64_t n = v.is_csr ? v.nelems_local : v.nrows_local;
-        return at::from_blob(v.base, {n, v.row_elems}, at::TensorOptions().dtype(v.st));
-    }
-
-    void epoch_begin() { fsm_.begin(); }
-    void epoch_end() { fsm_.end(); }
-
-    py::dict query(const std::string& name) {
-        HostVar& v = var(name);
-        py::dict d;
-        d["nrows_local"] = v.nrows_local;
-        d["nrows_total"] = v.prefix[nparts_];
-        d["disp"] = v.row_elems;
-        d["itemsize"] = v.itemsize;
-        d["is_csr"] = v.is_csr;
-        d["prefix"] = v.prefix;
-        if (v.is_csr) {
-            d["nelems_local"] = v.nelems_local;
-            d["elem_prefix"] = v.elem_prefix;
-        }
-        d["n_gather"] = v.n_gather;
-        d["rows_gathered"] = v.rows_gathered;
-        d["bytes_gathered"] = v.bytes_gathered;
-        return d;
-    }
-
-    bool has(const std::string& name) const { return vars_.count(name) > 0; }
-
-    void free_var(const std::string& name) {
-        auto it = vars_.find(name);
-        TORCH_CHECK(it != vars_.end(), "ddstore: unknown variable '", name, "'");
-        release(it->second);
-        vars_.erase(it);
-    }
-
-    void free_all() {
-        for (auto& kv : vars_) release(kv.second);
-        vars_.clear();
-    }
-
-private:
-    void check_new(const std::string& name) {
-        TORCH_CHECK(!vars_.count(name), "ddstore: variable '", name, "' already exists");
-    }
-    HostVar& var(const std::string& name) {
-        auto it = vars_.find(name);
-        TORCH_CHECK(it != vars_.end(), "ddstore: unknown variable '", name, "'");
-        return it->second;
-    }
-    void check_peers(const HostVar& v) {
-        TORCH_CHECK(!v.peers.empty(),
-                    "ddstore: variable not finalized (open_peers not called)");
-    }
-
-    void create_shm(HostVar& v, const std::string& name, size_t bytes) {
-        v.shm_name = shm_name(name);
-        TORCH_CHECK(v.shm_name.size() < 250, "ddstore: variable name too long");
-        v.base_bytes = bytes < 1 ? 1 : bytes;
-        shm_unlink(v.shm_name.c_str());  // stale segment from a crashed run
-        int fd = shm_open(v.shm_name.c_str(), O_CREAT | O_EXCL | O_RDWR, 0600);
-        TORCH_CHECK(fd >= 0, "ddstore: shm_open(create) failed for ", v.shm_name);
-        int rc = ftruncate(fd, (off_t)v.base_bytes);
-        if (rc != 0) {
-            close(fd);
-            shm_unlink(v.shm_name.c_str());
-            TORCH_CHECK(false, "ddstore: ftruncate failed (shm too large?)");
-        }
-        v.base = mmap(nullptr, v.base_bytes, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
-        close(fd);
-        TORCH_CHECK(v.base != MAP_FAILED, "ddstore: mmap failed");
-        v.creator = true;
-    }
-
-    void release(HostVar& v) {
-        for (size_t r = 0; r < v.peers.size(); ++r)
-            if (v.peer_bytes[r] > 0 && v.peers[r]) munmap(v.peers[r], v.peer_bytes[r]);
-        v.peers.clear();
-        if (v.base && v.base != MAP_FAILED) munmap(v.base, v.base_bytes);
-        if (v.creator) shm_unlink(v.shm_name.c_str());
-        v.base = nullptr;
-        v.active = false;
-    }
-
-    std::string session_;
-    int rank_;
-    int nparts_;
-    EpochFSM fsm_;
-    std::map<std::string, HostVar> vars_;
-};
-
-} // namespace
+using ddstore::DeviceStore;
+using ddstore::HostStore;
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "ddstore_amd native core (MI355X / gfx950)";
