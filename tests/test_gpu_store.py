@@ -360,3 +360,26 @@ def test_prefetch_csr_gpu(store):
             assert seg.shape[0] == lengths[g] and (seg == float(g)).all()
         seen += len(off_h) - 1
     assert seen == 300
+
+
+def test_pyddstore_numpy_staging_on_gpu():
+    # the reference-compatible NumPy surface backed by HBM shards
+    import pyddstore
+
+    p = pyddstore.PyDDStore()
+    try:
+        assert p.store.mode == "hip"
+        arr = np.random.rand(64, 8)
+        p.add("t", arr)
+        out = np.zeros((5, 8))
+        p.epoch_begin()
+        p.get("t", out, start=10)
+        p.epoch_end()
+        assert np.array_equal(out, arr[10:15])
+        p.init("u", 16, 4, itemsize=8)
+        p.update("u", np.full((4, 4), 2.0), offset=3)
+        ou = np.zeros((16, 4))
+        p.get("u", ou, 0)
+        assert (ou[3:7] == 2).all() and ou.sum() == 2.0 * 16
+    finally:
+        p.free()
