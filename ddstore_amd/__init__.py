@@ -11,7 +11,7 @@ from .store import DDStore
 from .distdataset import DistDataset, nsplit
 from .prefetch import PrefetchLoader
 from .reshuffle import reshuffle_epoch
-from . import io
+from . import debug, io
 
 __version__ = "0.1.0"
 
@@ -23,6 +23,7 @@ __all__ = [
     "nsplit",
     "PrefetchLoader",
     "reshuffle_epoch",
+    "debug",
     "io",
     "__version__",
 ]

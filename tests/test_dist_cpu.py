@@ -368,3 +368,29 @@ def test_ingest_from_files(tmp_path):
     np.save(tmp_path / "cl.npy", lens)
     np.save(tmp_path / "cv.npy", rng.random((int(lens.sum()), 2)).astype(np.float64))
     run_dist(_w_ingest_files, 2, str(tmp_path))
+
+
+# --------------------------------------------------------------------------
+def _w_verify_transport(rank, world):
+    from ddstore_amd import DDStore
+    from ddstore_amd.debug import verify_transport
+
+    s = DDStore(device="cpu")
+    arr = np.random.default_rng(rank).random((100, 8)).astype(np.float32)
+    s.add("x", arr)
+    r = verify_transport(s, "x", chunk_rows=33)
+    assert r["ok"], r
+    # corrupt the local shard AFTER digests would have matched -> re-verify
+    s.local_shard("x")[0, 0] += 1.0
+    r2 = verify_transport(s, "x")
+    # every reader (incl. self) should now flag this rank as mismatched...
+    # NB: owner digest is recomputed fresh, so corruption moves digest AND
+    # data together -> still consistent. Instead corrupt what readers see by
+    # checking digests differ from the first run:
+    assert r2["ok"]  # internally consistent again
+    assert r2["digests"][rank] != r["digests"][rank]
+    s.free()
+
+
+def test_verify_transport_ws3():
+    run_dist(_w_verify_transport, 3)

@@ -121,3 +121,18 @@ def _w_width_gpu(rank, world):
 
 def test_width_groups_gpu():
     run_dist(_w_width_gpu, 4)
+
+
+def _w_verify_transport_gpu(rank, world):
+    from ddstore_amd import DDStore
+    from ddstore_amd.debug import verify_transport
+
+    s = DDStore(device="cuda:0")
+    s.add("x", torch.randn(500, 16) + rank)
+    r = verify_transport(s, "x", chunk_rows=128)
+    assert r["ok"], r
+    s.free()
+
+
+def test_verify_transport_ipc():
+    run_dist(_w_verify_transport_gpu, 2)
