@@ -95,7 +95,12 @@ class Comm:
 
     def Split(self, color: int, key: int = 0) -> "Comm":  # noqa: N802
         """MPI_Comm_split semantics (reference distdataset.py:28 uses
-        ``comm.Split(rank // ddstore_width, rank)`` for replication groups)."""
+        ``comm.Split(rank // ddstore_width, rank)`` for replication groups).
+
+        Limitation vs MPI: torch.distributed groups always order members by
+        ascending global rank, so a ``key`` that would REORDER members
+        cannot be honored and raises (the reference's own usage, key=rank,
+        is order-preserving)."""
         if self._size == 1:
             return self
         triples = self.allgather((int(color), int(key), self._rank))
@@ -105,6 +110,11 @@ class Comm:
             members = sorted(
                 [(k, r) for cc, k, r in triples if cc == c]
             )  # order by (key, rank)
+            if [r for _, r in members] != sorted(r for _, r in members):
+                raise NotImplementedError(
+                    "ddstore Comm.Split: torch.distributed orders group members "
+                    "by ascending rank; a reordering `key` is not supported"
+                )
             ranks = [self._global_ranks[r] for _, r in members]
             # every rank must call new_group for every group, same order
             g = dist.new_group(ranks=ranks)
