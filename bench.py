@@ -126,18 +126,20 @@ def main():
     ntotal = rows * world
     nsteps_total = args.warmup + args.steps
     # sampler: per-epoch global permutation, this rank's disjoint slice
-    # (DistributedSampler semantics, reference vae-ddp.py:216)
+    # (DistributedSampler semantics, reference vae-ddp.py:216). The index
+    # pool is capped at 64 Mi rows (512 MB int64) and cycled for very long
+    # soaks -- far beyond any cache (64 Mi x 512 B = 32 GiB of rows), so
+    # reuse cannot make the fetch cheaper.
     g = torch.Generator().manual_seed(1234)
-    need = nsteps_total * batch
+    need = min(nsteps_total * batch, 64 * 1024 * 1024)
+    need = max(need, batch)
     order = []
     got = 0
-    epoch = 0
     while got < need:
         perm = torch.randperm(ntotal, generator=g)
         share = perm[rank::world]  # this rank's slice of the global shuffle
         order.append(share)
         got += share.numel()
-        epoch += 1
     order = torch.cat(order)[:need]
 
     store.epoch_begin()
@@ -165,9 +167,10 @@ def main():
             torch.empty(batch, dim, dtype=buf_dtype, device=device)
             for _ in range(nring)
         ]
+        nslots = order_dev.numel() // batch
         step_idx = [
             order_dev[k * batch : (k + 1) * batch].contiguous()
-            for k in range(nsteps_total)
+            for k in range(nslots)
         ]
         # one validated get_batch primes dtype/shape checks; the GPU loop
         # then uses the minimal-overhead gather_into path (the CPU compat
@@ -182,12 +185,10 @@ def main():
         def run_steps(n: int):
             k = counter["k"]
             for _ in range(n):
-                fetch_one("bench", step_idx[k], bufs[k % nring])
+                fetch_one("bench", step_idx[k % nslots], bufs[k % nring])
                 k += 1
             counter["k"] = k
 
-    sdt = {"f32": torch.float32, "bf16": torch.bfloat16, "u8": torch.uint8,
-           "fp8": torch.float8_e4m3fn}[args.store_dtype]
     if args.mode == "csr":
         # variable-length fetch: capacity ring buffers sized for the worst
         # batch (2*dim elems/sample max), gather_csr per step, no host sync
@@ -195,16 +196,17 @@ def main():
         nring = 4
         cap = batch * 2 * dim
         bufs = [torch.empty(cap, 1, dtype=torch.float32, device=tdev) for _ in range(nring)]
+        nslots = order_dev.numel() // batch
         step_idx = [
             order_dev[k * batch : (k + 1) * batch].contiguous()
-            for k in range(nsteps_total)
+            for k in range(nslots)
         ]
         counter = {"k": 0}
 
         def run_steps(n: int):
             k = counter["k"]
             for _ in range(n):
-                store.get_csr("bench", step_idx[k], out=bufs[k % nring])
+                store.get_csr("bench", step_idx[k % nslots], out=bufs[k % nring])
                 k += 1
             counter["k"] = k
 
