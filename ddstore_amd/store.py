@@ -382,6 +382,24 @@ class DDStore:
         self._backend.epoch_end()
         self.comm.barrier()
 
+    def epoch(self):
+        """Context manager form of the epoch fence:
+
+            with store.epoch():
+                batch = store.get_batch(...)
+        """
+        import contextlib
+
+        @contextlib.contextmanager
+        def _cm():
+            self.epoch_begin()
+            try:
+                yield self
+            finally:
+                self.epoch_end()
+
+        return _cm()
+
     # --------------------------------------------------------------- reshuffle
     def reshuffle(self, name: str, seed: int) -> None:
         """Epoch-level global data reshuffle over xGMI (all-to-all); see

@@ -327,3 +327,19 @@ def test_prefetch_ragged_tail(store):
     store.add("rt", arr)
     sizes = [b.shape[0] for b in PrefetchLoader(store, "rt", np.arange(50), 16)]
     assert sizes == [16, 16, 16, 2]
+
+
+def test_epoch_context_manager(store):
+    store.add("ec", np.ones((4, 2), dtype=np.float32))
+    with store.epoch():
+        out = store.get_batch("ec", [1])
+        assert out[0, 0] == 1.0
+    # closed: a fresh begin works
+    store.epoch_begin()
+    store.epoch_end()
+    # exceptions still close the epoch
+    with pytest.raises(RuntimeError):
+        with store.epoch():
+            store.get("ec", np.zeros((9, 2), dtype=np.float32), 0)  # invalid count
+    store.epoch_begin()
+    store.epoch_end()
