@@ -65,6 +65,8 @@ class PrefetchLoader:
         self.indices = idx
         self.is_csr = store._meta(name)["is_csr"]
         if self.is_csr:
+            if affine is not None:
+                raise ValueError("affine is not supported for CSR variables")
             if label_name is not None and store._meta(label_name)["is_csr"]:
                 raise ValueError("label variable must be fixed-stride")
             # capacity ring buffers sized for the worst batch
