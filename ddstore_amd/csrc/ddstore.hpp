@@ -254,7 +254,12 @@ public:
                 hipIpcMemHandle_t h;
                 std::memcpy(&h, handles[r].data(), sizeof(h));
                 void* p = nullptr;
-                HIP_CHECK(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess));
+                hipError_t e = hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess);
+                TORCH_CHECK(e == hipSuccess,
+                            "ddstore open_peers: hipIpcOpenMemHandle failed for peer rank ",
+                            r, " (", hipGetErrorString(e),
+                            ") -- peers must share one node/xGMI domain and "
+                            "HSA_ENABLE_IPC_MODE_LEGACY=0 must be set");
                 v.peers[r] = p;
                 v.opened[r] = 1;
             }
