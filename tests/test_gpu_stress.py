@@ -67,3 +67,14 @@ def test_no_leak_across_store_lifecycles():
     # torch's caching allocator may retain some blocks; raw shard leaks would
     # show as ~30 x 16 MiB = 480 MiB
     assert leaked < 200 * 2**20, f"leaked {leaked/2**20:.0f} MiB over 30 cycles"
+
+
+def test_fuzz_shadow_model_gpu():
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", "tools"))
+    from fuzz_store import run as fuzz_run
+
+    n = fuzz_run(ops=250, seed=7, device="cuda:0")
+    assert n > 30

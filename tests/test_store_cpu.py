@@ -343,3 +343,12 @@ def test_epoch_context_manager(store):
             store.get("ec", np.zeros((9, 2), dtype=np.float32), 0)  # invalid count
     store.epoch_begin()
     store.epoch_end()
+
+
+def test_fuzz_shadow_model():
+    import sys, os
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", "tools"))
+    from fuzz_store import run as fuzz_run
+
+    n = fuzz_run(ops=150, seed=99, device="cpu")
+    assert n > 20
