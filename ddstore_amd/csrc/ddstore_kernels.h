@@ -77,13 +77,6 @@ void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
               int64_t nidx, int64_t nsamples, int64_t* d_lens,
               unsigned long long* d_oob);
 
-// Fused CSR plan: out_off[0]=0, out_off[i+1]=cumsum of sample lengths, in 3
-// small kernels (block_tot scratch: csr_plan_blocks(nidx) int64s). Returns
-// false when nidx exceeds the single-pass limit (caller falls back).
-bool csr_plan(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
-              int64_t nidx, int64_t nsamples, int64_t* d_out_off,
-              int64_t* d_block_tot, unsigned long long* d_oob);
-int64_t csr_plan_blocks(int64_t nidx);
 
 // Scatter rows of a packed buffer into the local shard at arbitrary local row
 // ids (inverse of gather_rows with nparts==1). Used by the epoch reshuffle to

@@ -315,92 +315,6 @@ k_gather_csr_wave(const void* const* peer_base,
     }
 }
 
-// ---------------------------------------------------------------------------
-// Fused CSR plan scan: out_off[i+1] = sum of lens(idx[0..i]) in 3 small
-// kernels (lens computed inline) instead of zeros + lens + rocprim's
-// init+lookback dispatches. Block A covers 4096 samples (256 threads x 16):
-// per-thread serial scan + block Hillis-Steele + per-block total; B scans
-// the <=4096 block totals in one block; C adds block bases.
-// ---------------------------------------------------------------------------
-constexpr int kScanT = 256;
-constexpr int kScanPerT = 16;
-constexpr int64_t kScanBlock = (int64_t)kScanT * kScanPerT;  // 4096 samples
-constexpr int64_t kScanMaxN = kScanBlock * kScanBlock;       // 16.7M samples
-
-__global__ void __launch_bounds__(kScanT)
-k_csr_plan_a(const int64_t* goff, const int64_t* idx, int64_t nidx,
-             int64_t nsamples, int64_t* __restrict__ out_off,
-             int64_t* __restrict__ block_tot, unsigned long long* oob) {
-    __shared__ int64_t sh[kScanT];
-    const int t = threadIdx.x;
-    const int64_t base = (int64_t)blockIdx.x * kScanBlock;
-    int64_t local_excl[kScanPerT];
-    int64_t sum = 0;
-#pragma unroll
-    for (int j = 0; j < kScanPerT; ++j) {
-        const int64_t i = base + (int64_t)t * kScanPerT + j;
-        local_excl[j] = sum;
-        if (i < nidx) {
-            const int64_t g = idx[i];
-            if (g < 0 || g >= nsamples) {
-                atomicAdd(oob, 1ull);
-            } else {
-                sum += goff[g + 1] - goff[g];
-            }
-        }
-    }
-    sh[t] = sum;
-    __syncthreads();
-    for (int off = 1; off < kScanT; off <<= 1) {
-        const int64_t v = (t >= off) ? sh[t - off] : 0;
-        __syncthreads();
-        sh[t] += v;
-        __syncthreads();
-    }
-    const int64_t thread_base = (t > 0) ? sh[t - 1] : 0;
-#pragma unroll
-    for (int j = 0; j < kScanPerT; ++j) {
-        const int64_t i = base + (int64_t)t * kScanPerT + j;
-        if (i < nidx) out_off[i + 1] = thread_base + local_excl[j];
-    }
-    if (t == kScanT - 1) block_tot[blockIdx.x] = sh[kScanT - 1];
-    if (blockIdx.x == 0 && t == 0) out_off[0] = 0;
-}
-
-__global__ void __launch_bounds__(kScanT)
-k_csr_plan_b(int64_t* __restrict__ block_tot, int nblocks) {
-    // single block: in-place EXCLUSIVE scan of block totals
-    __shared__ int64_t sh[kScanT];
-    const int t = threadIdx.x;
-    int64_t running = 0;
-    for (int chunk = 0; chunk < nblocks; chunk += kScanT) {
-        const int i = chunk + t;
-        const int64_t v = (i < nblocks) ? block_tot[i] : 0;
-        sh[t] = v;
-        __syncthreads();
-        for (int off = 1; off < kScanT; off <<= 1) {
-            const int64_t w = (t >= off) ? sh[t - off] : 0;
-            __syncthreads();
-            sh[t] += w;
-            __syncthreads();
-        }
-        const int64_t incl = sh[t];
-        const int64_t chunk_total = sh[kScanT - 1];
-        if (i < nblocks) block_tot[i] = running + (incl - v);
-        running += chunk_total;
-        __syncthreads();  // sh reused next chunk
-    }
-}
-
-__global__ void __launch_bounds__(kBlock)
-k_csr_plan_c(int64_t* __restrict__ out_off, const int64_t* __restrict__ block_tot,
-             int64_t nidx) {
-    for (int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x; i < nidx;
-         i += (int64_t)gridDim.x * kBlock) {
-        out_off[i + 1] += block_tot[i / kScanBlock];
-    }
-}
-
 // Per-sample lengths for a CSR gather plan: lens[i] = goff[idx[i]+1] -
 // goff[idx[i]] (one kernel instead of the 3-4 elementwise torch launches
 // the equivalent `goff[idx+1]-goff[idx]` costs per step).
@@ -698,25 +612,6 @@ void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
                        d_goff, d_idx, nidx, nsamples, d_lens, d_oob);
 }
 
-bool csr_plan(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
-              int64_t nidx, int64_t nsamples, int64_t* d_out_off,
-              int64_t* d_block_tot, unsigned long long* d_oob) {
-    if (nidx == 0 || nidx > kScanMaxN) return false;
-    const int nb = (int)((nidx + kScanBlock - 1) / kScanBlock);
-    hipLaunchKernelGGL(k_csr_plan_a, dim3(nb), dim3(kScanT), 0, stream,
-                       d_goff, d_idx, nidx, nsamples, d_out_off, d_block_tot, d_oob);
-    if (nb > 1) {
-        hipLaunchKernelGGL(k_csr_plan_b, dim3(1), dim3(kScanT), 0, stream,
-                           d_block_tot, nb);
-        hipLaunchKernelGGL(k_csr_plan_c, dim3(n_blocks(nidx)), dim3(kBlock), 0,
-                           stream, d_out_off, d_block_tot, nidx);
-    }
-    return true;
-}
-
-int64_t csr_plan_blocks(int64_t nidx) {
-    return (nidx + kScanBlock - 1) / kScanBlock;
-}
 
 void gather_csr(hipStream_t stream,
                 const void* const* d_peer_base,
