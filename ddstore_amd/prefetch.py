@@ -8,6 +8,14 @@ for sample fetch (BASELINE config 5). Synchronization is via HIP events:
 
     side stream:  [wait buf free] -> gather batch k -> record ready_k
     main stream:  wait ready_k -> train step on batch k -> record free_k
+
+Allocator note: the data/label ring buffers are allocated once and reused,
+so the caching allocator never recycles them mid-flight. CSR offset tensors
+ARE allocated per batch on the side stream; their memory can only be
+recycled by a later side-stream allocation whose writes are enqueued after
+``side.wait_event(free_k)``, and the consumer's reads of the old offsets
+were enqueued on the main stream before ``free_k`` was recorded -- so every
+reuse is event-ordered after every read.
 """
 from __future__ import annotations
 
