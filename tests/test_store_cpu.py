@@ -352,3 +352,26 @@ def test_fuzz_shadow_model():
 
     n = fuzz_run(ops=150, seed=99, device="cpu")
     assert n > 20
+
+
+def test_io_single_rank(tmp_path, store):
+    from ddstore_amd import io as dio
+
+    arr = np.random.rand(30, 5).astype(np.float64)
+    np.save(tmp_path / "a.npy", arr)
+    dio.add_from_npy(store, "a", str(tmp_path / "a.npy"))
+    out = store.get_batch("a", list(range(30)))
+    assert np.array_equal(out.numpy(), arr)
+    raw = np.random.rand(12, 6).astype(np.float32)
+    raw.tofile(tmp_path / "b.bin")
+    dio.add_from_memmap(store, "b", str(tmp_path / "b.bin"), np.float32, (6,))
+    assert store.query("b")["nrows_total"] == 12
+    with pytest.raises(ValueError, match="whole number of rows"):
+        dio.add_from_memmap(store, "bad", str(tmp_path / "b.bin"), np.float32, (7,))
+
+
+def test_variables_listing(store):
+    assert store.variables() == []
+    store.add("v1", np.zeros((2, 2), dtype=np.float32))
+    store.add_csr("v2", np.zeros((3, 1), dtype=np.float32), [1, 2])
+    assert sorted(store.variables()) == ["v1", "v2"]
