@@ -56,7 +56,18 @@ def _as_tensor(arr: ArrayLike) -> torch.Tensor:
     if isinstance(arr, torch.Tensor):
         t = arr
     elif isinstance(arr, np.ndarray):
-        t = torch.from_numpy(np.ascontiguousarray(arr))
+        a = np.ascontiguousarray(arr)
+        if not a.flags.writeable:
+            # read-only views (e.g. io.py memmaps) are only ever READ here
+            # (add/update copy into store-owned memory); silence torch's
+            # non-writable warning instead of paying a host copy
+            import warnings
+
+            with warnings.catch_warnings():
+                warnings.simplefilter("ignore", UserWarning)
+                t = torch.from_numpy(a)
+        else:
+            t = torch.from_numpy(a)
     else:
         t = torch.as_tensor(arr)
     if t.dtype not in _SUPPORTED:
