@@ -117,6 +117,25 @@ def main():
     avg_total = sum(p[2] for p in plans) / NIDX
     report(f"gather_csr ~1KiB samples (x{ncsr})", el, avg_total * 8)
 
+    # dense get_range (the reference's core read primitive): big contiguous
+    # block, device-to-device through the peer pointer path
+    rout = torch.empty(1 << 16, dim, device=dev)
+    el = timeit(lambda i: s._backend.get_range("f32", (i % 8) << 16, 1 << 16, rout))
+    report("get_range 32 MiB dense block (D2D)", el, (1 << 16) * dim * 8)
+
+    # reference demo workload: single 512 B row per get, synchronous to host
+    # (test/demo.py:45-50 pattern) -- per-op latency, not bandwidth
+    import numpy as np2
+    one = np2.zeros((1, dim), dtype=np2.float32)
+    t0 = time.perf_counter()
+    NGET = 200
+    for i in range(NGET):
+        s.get("f32", one, start=(i * 7919) % rows)
+    el = (time.perf_counter() - t0) / NGET
+    if not args.json:
+        print(f"{'compat get() 1 row, sync D2H':38s} {el*1e6:8.1f} us/get")
+    results["compat_get_1row_us"] = {"us": el * 1e6, "GBps": dim * 4 / el / 1e9}
+
     s.free()
     if args.json:
         print(json.dumps(results))
