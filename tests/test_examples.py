@@ -37,3 +37,14 @@ def test_gnn_example():
     out = _run(["examples/gnn_csr_train.py", "--epochs", "1",
                 "--graphs-per-rank", "1500", "--device", "cpu"])
     assert "acc" in out
+
+
+def test_demo_example_torchrun_ws2():
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+         "examples/demo.py", "--num", "3000", "--nbatch", "4", "--device", "cpu"],
+        cwd=ROOT, capture_output=True, text=True, timeout=280,
+    )
+    assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+    assert "verified" in r.stdout

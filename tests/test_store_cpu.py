@@ -375,3 +375,18 @@ def test_variables_listing(store):
     store.add("v1", np.zeros((2, 2), dtype=np.float32))
     store.add_csr("v2", np.zeros((3, 1), dtype=np.float32), [1, 2])
     assert sorted(store.variables()) == ["v1", "v2"]
+
+
+def test_dump_load_after_reshuffle(tmp_path, store):
+    from ddstore_amd.reshuffle import expected_perm
+
+    arr = np.arange(40, dtype=np.float32).reshape(20, 2)
+    store.add("dr", arr)
+    store.reshuffle("dr", seed=3)
+    store.dump("dr", str(tmp_path / "dr.pt"))
+    s2 = DDStore(device="cpu")
+    s2.load("dr", str(tmp_path / "dr.pt"))
+    perm = expected_perm(20, 3, store.device).numpy()
+    out = s2.get_batch("dr", list(range(20)))
+    assert np.array_equal(out.numpy(), arr[perm])
+    s2.free()
