@@ -27,7 +27,10 @@ from ddstore_amd import DDStore  # noqa: E402
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--num", type=int, default=1024 * 1024)
+    # --rows alias: under torchrun, bare --num is ambiguous with
+    # torchrun's own --numa-binding (argparse prefix matching)
+    p.add_argument("--num", "--rows", dest="num", type=int,
+                   default=1024 * 1024)
     p.add_argument("--dim", type=int, default=64)
     p.add_argument("--nbatch", type=int, default=32)
     p.add_argument("--device", default=None)

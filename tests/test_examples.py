@@ -43,7 +43,7 @@ def test_demo_example_torchrun_ws2():
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--standalone",
          "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
-         "examples/demo.py", "--num", "3000", "--nbatch", "4", "--device", "cpu"],
+         "examples/demo.py", "--rows", "3000", "--nbatch", "4", "--device", "cpu"],
         cwd=ROOT, capture_output=True, text=True, timeout=280,
     )
     assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
