@@ -488,6 +488,7 @@ public:
         fsm_.end();
         HIP_CHECK(hipStreamSynchronize(stream()));
     }
+    bool epoch_active() const { return fsm_.fence_active; }
 
     py::dict query(const std::string& name) {
         DeviceVar& v = var(name);
@@ -864,6 +865,7 @@ public:
 
     void epoch_begin() { fsm_.begin(); }
     void epoch_end() { fsm_.end(); }
+    bool epoch_active() const { return fsm_.fence_active; }
 
     py::dict query(const std::string& name) {
         HostVar& v = var(name);

@@ -52,6 +52,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("local_shard", &DeviceStore::local_shard)
         .def("epoch_begin", &DeviceStore::epoch_begin)
         .def("epoch_end", &DeviceStore::epoch_end)
+        .def("epoch_active", &DeviceStore::epoch_active)
         .def("query", &DeviceStore::query)
         .def("has", &DeviceStore::has)
         .def("free_var", &DeviceStore::free_var)
@@ -73,6 +74,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("local_shard", &HostStore::local_shard)
         .def("epoch_begin", &HostStore::epoch_begin)
         .def("epoch_end", &HostStore::epoch_end)
+        .def("epoch_active", &HostStore::epoch_active)
         .def("query", &HostStore::query)
         .def("has", &HostStore::has)
         .def("free_var", &HostStore::free_var)

@@ -25,6 +25,11 @@ from .store import DDStore
 
 
 def reshuffle_epoch(store: DDStore, name: str, seed: int) -> None:
+    if store._backend.epoch_active():
+        raise RuntimeError(
+            "ddstore reshuffle: cannot move data inside an open epoch "
+            "(concurrent gets would race); call epoch_end() first"
+        )
     q = store.query(name)
     if q["is_csr"]:
         return _reshuffle_csr(store, name, seed)

@@ -390,3 +390,12 @@ def test_dump_load_after_reshuffle(tmp_path, store):
     out = s2.get_batch("dr", list(range(20)))
     assert np.array_equal(out.numpy(), arr[perm])
     s2.free()
+
+
+def test_reshuffle_inside_epoch_rejected(store):
+    store.add("re", np.zeros((8, 4), dtype=np.float32))
+    store.epoch_begin()
+    with pytest.raises(RuntimeError, match="open epoch"):
+        store.reshuffle("re", seed=1)
+    store.epoch_end()
+    store.reshuffle("re", seed=1)  # fine outside
