@@ -234,7 +234,10 @@ class DDStore:
 
     def update(self, name: str, arr: ArrayLike, offset: int = 0) -> None:
         """Local fill at row ``offset`` -- no communication, no epoch required
-        (reference ddstore.hpp:181-195)."""
+        (reference ddstore.hpp:181-195). Like the reference, separating
+        updates from remote reads is the CALLER's job: write outside the
+        epoch windows in which peers read (the fence choreography of
+        vae-ddp.py:240-265 exists exactly for this)."""
         t = _as_tensor(arr)
         self._backend.update(name, self._staged(t), int(offset))
 
