@@ -83,10 +83,30 @@ def run(ops: int, seed: int, device=None, verbose: bool = False) -> int:
             arr = fixed[name]
             k = int(rng.integers(0, 300))
             idx = rng.integers(0, arr.shape[0], size=k)
-            out = store.get_batch(name, idx.astype(np.int64))
-            if store.mode == "hip":
-                torch.cuda.synchronize()
-            assert np.array_equal(out.cpu().numpy(), arr[idx]), (op, name)
+            mode = rng.integers(3)
+            if mode == 0:       # same dtype (byte move)
+                out = store.get_batch(name, idx.astype(np.int64))
+                if store.mode == "hip":
+                    torch.cuda.synchronize()
+                assert np.array_equal(out.cpu().numpy(), arr[idx]), (op, name)
+            elif mode == 1:     # fused cast, verified vs torch's own cast
+                ot = [torch.float32, torch.float64, torch.int64, torch.bfloat16,
+                      torch.float16][rng.integers(5)]
+                out = store.get_batch(name, idx.astype(np.int64), dtype=ot)
+                if store.mode == "hip":
+                    torch.cuda.synchronize()
+                ref = torch.from_numpy(arr[idx]).to(ot)
+                assert torch.equal(out.cpu(), ref), (op, name, ot)
+            else:               # fused affine
+                sc = float(rng.uniform(0.1, 3.0))
+                sh = float(rng.uniform(-2.0, 2.0))
+                out = store.get_batch(name, idx.astype(np.int64),
+                                      dtype=torch.float32, affine=(sc, sh))
+                if store.mode == "hip":
+                    torch.cuda.synchronize()
+                ref = (torch.from_numpy(arr[idx]).to(torch.float32) * sc + sh)
+                assert torch.allclose(out.cpu(), ref, atol=1e-4, rtol=1e-5), (
+                    op, name, sc, sh)
             nchecks += 1
         elif op == "update":
             name = list(fixed)[rng.integers(len(fixed))]
