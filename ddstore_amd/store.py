@@ -456,7 +456,6 @@ class DDStore:
         shard = payload["shard"]
         if payload["is_csr"]:
             goff = payload["goff"]
-            ep = payload["elem_prefix"]
             p = payload["prefix"]
             lo, hi = p[self.rank], p[self.rank + 1]
             lengths = (goff[lo + 1 : hi + 1] - goff[lo:hi]) if hi > lo else torch.zeros(
