@@ -48,6 +48,9 @@ def main():
     p.add_argument("--graphs-per-rank", type=int, default=20000)
     p.add_argument("--batch-size", type=int, default=512)
     p.add_argument("--device", default=None)
+    p.add_argument("--backend", default=None,
+                   help="dist backend override (default nccl on GPU; use gloo "
+                        "to oversubscribe ranks on one GPU)")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -56,11 +59,14 @@ def main():
     use_cuda = torch.cuda.is_available() if args.device is None else str(
         args.device).startswith("cuda")
     if world > 1:
-        dist.init_process_group("nccl" if use_cuda else "gloo",
-                                rank=rank, world_size=world)
-    device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
+        backend = args.backend or ("nccl" if use_cuda else "gloo")
+        dist.init_process_group(backend, rank=rank, world_size=world)
     if use_cuda:
+        local_rank = local_rank % max(torch.cuda.device_count(), 1)
+        device = torch.device("cuda", local_rank)
         torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
 
     store = DDStore(device=device if use_cuda else "cpu")
 

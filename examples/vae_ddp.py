@@ -75,6 +75,9 @@ def main():
     p.add_argument("--batch-size", type=int, default=128)
     p.add_argument("--nsamples", type=int, default=12000)
     p.add_argument("--device", default=None)
+    p.add_argument("--backend", default=None,
+                   help="dist backend override (default nccl on GPU; use gloo "
+                        "to oversubscribe ranks on one GPU)")
     p.add_argument("--ddstore-width", type=int, default=None)
     args = p.parse_args()
 
@@ -84,11 +87,14 @@ def main():
     use_cuda = torch.cuda.is_available() if args.device is None else str(
         args.device).startswith("cuda")
     if world > 1:
-        dist.init_process_group("nccl" if use_cuda else "gloo",
-                                rank=rank, world_size=world)
-    device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
+        backend = args.backend or ("nccl" if use_cuda else "gloo")
+        dist.init_process_group(backend, rank=rank, world_size=world)
     if use_cuda:
+        local_rank = local_rank % max(torch.cuda.device_count(), 1)
+        device = torch.device("cuda", local_rank)
         torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
 
     # synthetic MNIST: blobby per-class patterns so the VAE has structure to fit
     rng = np.random.default_rng(0)  # same dataset on every rank
