@@ -48,3 +48,29 @@ def test_demo_example_torchrun_ws2():
     )
     assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
     assert "verified" in r.stdout
+
+
+@pytest.mark.gpu
+def test_vae_example_ws2_gpu():
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+         "examples/vae_ddp.py", "--epochs", "1", "--nsamples", "3000",
+         "--backend", "gloo"],
+        cwd=ROOT, capture_output=True, text=True, timeout=280,
+    )
+    assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+    assert "train loss" in r.stdout
+
+
+@pytest.mark.gpu
+def test_gnn_example_ws2_gpu():
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+         "examples/gnn_csr_train.py", "--epochs", "1",
+         "--graphs-per-rank", "4000", "--backend", "gloo"],
+        cwd=ROOT, capture_output=True, text=True, timeout=280,
+    )
+    assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+    assert "acc" in r.stdout
