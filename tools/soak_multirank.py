@@ -69,7 +69,7 @@ def main():
             assert r["ok"], r
         if use_cuda:
             torch.cuda.synchronize()
-        assert store.query("x")["oob_skipped"] == 0
+        assert store.query("x").get("oob_skipped", 0) == 0
         cycles += 1
     if rank == 0:
         print(f"soak OK: world={world} {cycles} cycles, "
