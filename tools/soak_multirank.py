@@ -25,7 +25,12 @@ from ddstore_amd.debug import verify_transport  # noqa: E402
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--seconds", type=float, default=120)
-    p.add_argument("--gib", type=float, default=2.0, help="fixed shard GiB/rank")
+    p.add_argument("--gib", type=float, default=0.5, help="fixed shard GiB/rank")
+    p.add_argument("--reshuffle-every", type=int, default=1,
+                   help="reshuffle cadence in cycles (0 = never). NB: with a "
+                        "gloo backend the all-to-all stages through CPU/TCP "
+                        "and is VERY slow for multi-GiB shards; use nccl on "
+                        "real multi-GPU nodes or keep shards small")
     p.add_argument("--backend", default=None)
     p.add_argument("--device", default=None)
     args = p.parse_args()
@@ -63,7 +68,8 @@ def main():
             cidx = torch.randint(0, ncsr, (65536,),
                                  generator=g)
             v, off = store.get_csr("c", cidx)
-        store.reshuffle("x", seed=cycles)
+        if args.reshuffle_every and cycles % args.reshuffle_every == 0:
+            store.reshuffle("x", seed=cycles)
         if cycles % 5 == 0:
             r = verify_transport(store, "x")
             assert r["ok"], r
@@ -71,6 +77,9 @@ def main():
             torch.cuda.synchronize()
         assert store.query("x").get("oob_skipped", 0) == 0
         cycles += 1
+        if rank == 0 and cycles % 5 == 0:
+            print(f"  cycle {cycles}, {time.time() - (t_end - args.seconds):.0f}s",
+                  flush=True)
     if rank == 0:
         print(f"soak OK: world={world} {cycles} cycles, "
               f"{fetched/1e6:.0f}M rows fetched/rank, verifier clean")
