@@ -61,8 +61,9 @@ def main():
     while time.time() < t_end:
         g = torch.Generator().manual_seed(cycles)
         mine = torch.randperm(ntotal, generator=g)[rank::world][: 2**20]
+        bsz = max(1024, min(131072, mine.numel() // 4))
         with store.epoch():
-            for b in PrefetchLoader(store, "x", mine, 131072,
+            for b in PrefetchLoader(store, "x", mine, bsz,
                                     out_dtype=torch.bfloat16, drop_last=True):
                 fetched += b.shape[0]
             cidx = torch.randint(0, ncsr, (65536,),
