@@ -102,3 +102,13 @@ def _w_split_keys(rank, world):
 
 def test_split_key_ordering():
     run_dist(_w_split_keys, 3)
+
+
+def _w_bcast_nonzero_root(rank, world):
+    c = Comm()
+    v = c.bcast("from2" if rank == 2 else None, root=2)
+    assert v == "from2"
+
+
+def test_bcast_nonzero_root():
+    run_dist(_w_bcast_nonzero_root, 3)

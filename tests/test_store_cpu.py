@@ -399,3 +399,29 @@ def test_reshuffle_inside_epoch_rejected(store):
         store.reshuffle("re", seed=1)
     store.epoch_end()
     store.reshuffle("re", seed=1)  # fine outside
+
+
+def test_device_arg_forms():
+    # int / "cpu" device argument forms
+    s = DDStore(device="cpu")
+    assert s.mode == "shm"
+    s.free()
+
+
+def test_init_explicit_dtype(store):
+    store.init("ex", 6, 3, dtype=torch.int64)
+    store.update("ex", np.arange(6, dtype=np.int64).reshape(2, 3), offset=1)
+    out = store.get_batch("ex", [0, 1, 2])
+    assert out.dtype == torch.int64
+    assert out[0].sum() == 0 and out[1].tolist() == [0, 1, 2]
+
+
+def test_prefetch_len_and_drop_last(store):
+    from ddstore_amd import PrefetchLoader
+
+    store.add("pl", np.zeros((50, 2), dtype=np.float32))
+    keep = PrefetchLoader(store, "pl", np.arange(50), 16)
+    drop = PrefetchLoader(store, "pl", np.arange(50), 16, drop_last=True)
+    assert len(keep) == 4 and len(drop) == 3
+    assert sum(b.shape[0] for b in keep) == 50
+    assert sum(b.shape[0] for b in drop) == 48
