@@ -24,6 +24,13 @@ def _dist_ready() -> bool:
     return dist.is_available() and dist.is_initialized()
 
 
+# Process-lifetime cache of created groups keyed by their member ranks:
+# torch.distributed groups are never garbage-collected implicitly, so a
+# repeated Split (e.g. many DDStore(ddstore_width=...) constructions in one
+# process) must reuse instead of leaking a new group per call.
+_group_cache: dict = {}
+
+
 class Comm:
     """Collective metadata plane over torch.distributed (or a no-op self comm)."""
 
@@ -116,8 +123,14 @@ class Comm:
                     "by ascending rank; a reordering `key` is not supported"
                 )
             ranks = [self._global_ranks[r] for _, r in members]
-            # every rank must call new_group for every group, same order
-            g = dist.new_group(ranks=ranks)
+            key = tuple(ranks)
+            g = _group_cache.get(key)
+            if g is None:
+                # every rank must call new_group for every group, same order;
+                # the cache is keyed identically on every rank so cache
+                # hits/misses agree and the collective stays matched
+                g = dist.new_group(ranks=ranks)
+                _group_cache[key] = g
             if c == color:
                 my_group = g
         return Comm(my_group)

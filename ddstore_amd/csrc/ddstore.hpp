@@ -18,6 +18,8 @@
 #include <sys/mman.h>
 #include <unistd.h>
 
+#include <atomic>
+#include <cstdlib>
 #include <cstring>
 #include <map>
 #include <string>
@@ -84,6 +86,18 @@ inline int owner_of_host(const std::vector<int64_t>& prefix, int64_t row) {
         if (prefix[mid] <= row) lo = mid; else hi = mid - 1;
     }
     return lo;
+}
+
+// DDSTORE_STRICT=1: out-of-range / over-capacity samples raise at the next
+// host sync point (epoch_end) instead of being silently skipped+counted --
+// the reference THROWS on a bad get (reference ddstore.hpp:210-214); the
+// default here keeps the async hot path and counts (query()["oob_skipped"]).
+inline bool strict_mode() {
+    static const bool s = [] {
+        const char* e = std::getenv("DDSTORE_STRICT");
+        return e != nullptr && e[0] == '1';
+    }();
+    return s;
 }
 
 struct EpochFSM {
@@ -229,6 +243,65 @@ public:
         vars_[name] = std::move(v);
     }
 
+    void init_csr(const std::string& name, int64_t nsamples, int64_t nelems,
+                  int64_t row_elems, at::ScalarType st,
+                  std::vector<int64_t> nsamples_all,
+                  std::vector<int64_t> nelems_all, const at::Tensor& goff_cpu) {
+        // CSR analog of init: lengths are fixed at registration (they define
+        // the global offset directory), VALUES are zeroed and filled later
+        // with update_elems -- the incremental-fill pattern the reference
+        // supports for fixed-stride variables (ddstore.hpp:110-195,
+        // README.md:107), extended to the first-class CSR layout.
+        check_new(name);
+        at::Tensor proto = at::empty({0}, at::TensorOptions().dtype(st));
+        DeviceVar v;
+        v.is_csr = true;
+        v.st = st;
+        v.dds_t = dds_type_of(proto);
+        v.itemsize = dds_itemsize(v.dds_t);
+        v.row_elems = row_elems;
+        v.nrows_local = nsamples;
+        v.nelems_local = nelems;
+        v.prefix = make_prefix(nsamples_all);
+        v.elem_prefix = make_prefix(nelems_all);
+        TORCH_CHECK(goff_cpu.scalar_type() == at::kLong && goff_cpu.is_contiguous() &&
+                        goff_cpu.device().is_cpu() &&
+                        goff_cpu.numel() == v.prefix[nparts_] + 1,
+                    "ddstore init_csr: bad global offsets");
+        alloc_base(v, (size_t)(nelems * row_elems * v.itemsize));
+        try {
+            HIP_CHECK(hipMemsetAsync(v.base, 0, v.base_bytes, stream()));
+            HIP_CHECK(hipStreamSynchronize(stream()));
+            size_t gbytes = (size_t)goff_cpu.numel() * 8;
+            HIP_CHECK(hipMalloc((void**)&v.d_goff, gbytes));
+            HIP_CHECK(hipMemcpy(v.d_goff, goff_cpu.data_ptr<int64_t>(), gbytes,
+                                hipMemcpyHostToDevice));
+        } catch (...) {
+            release(v);
+            throw;
+        }
+        v.active = true;
+        vars_[name] = std::move(v);
+    }
+
+    void update_elems(const std::string& name, const at::Tensor& src,
+                      int64_t elem_offset) {
+        // purely local CSR fill: elements [elem_offset, elem_offset+k) of
+        // this rank's shard (element offsets are LOCAL; the Python layer
+        // translates a sample offset via the goff directory)
+        DeviceVar& v = var(name);
+        TORCH_CHECK(v.is_csr, "ddstore update_elems: CSR variables only");
+        TORCH_CHECK(src.is_contiguous(), "ddstore update: array must be C-contiguous");
+        TORCH_CHECK(dds_itemsize(dds_type_of(src)) == v.itemsize,
+                    "ddstore update: itemsize mismatch");
+        int64_t nelems = src.numel() / std::max<int64_t>(v.row_elems, 1);
+        TORCH_CHECK(src.numel() == nelems * v.row_elems,
+                    "ddstore update: shape mismatch");
+        TORCH_CHECK(elem_offset >= 0 && elem_offset + nelems <= v.nelems_local,
+                    "ddstore update: out of range");
+        ingest(v.base, src, elem_offset * v.row_elems * v.itemsize);
+    }
+
     py::bytes ipc_handle(const std::string& name) {
         DeviceVar& v = var(name);
         hipIpcMemHandle_t h;
@@ -264,9 +337,9 @@ public:
                 v.opened[r] = 1;
             }
         }
-        // one device block: [peer ptrs | prefix | elem_prefix? | oob counter]
+        // one device block: [peer ptrs | prefix | elem_prefix? | counters]
         size_t nb = nparts_ * sizeof(void*) + (nparts_ + 1) * 8 +
-                    (v.is_csr ? (nparts_ + 1) * 8 : 0) + 8;
+                    (v.is_csr ? (nparts_ + 1) * 8 : 0) + DDS_NCTR * 8;
         HIP_CHECK(hipMalloc((void**)&v.d_peers, nb));
         v.d_prefix = reinterpret_cast<int64_t*>(v.d_peers + nparts_);
         HIP_CHECK(hipMemcpy(v.d_peers, v.peers.data(), nparts_ * sizeof(void*),
@@ -281,7 +354,7 @@ public:
             tail += nparts_ + 1;
         }
         v.d_oob = reinterpret_cast<unsigned long long*>(tail);
-        HIP_CHECK(hipMemset(v.d_oob, 0, 8));
+        HIP_CHECK(hipMemset(v.d_oob, 0, DDS_NCTR * 8));
     }
 
     void update(const std::string& name, const at::Tensor& src, int64_t offset) {
@@ -394,20 +467,27 @@ public:
                     "ddstore gather_csr: bad output tensor");
         TORCH_CHECK(out.numel() >= total_elems * v.row_elems,
                     "ddstore gather_csr: output too small");
+        // capacity passed to the kernel is the REAL buffer capacity: a sample
+        // whose slice would end past it is skipped + counted, never written
+        // out of bounds
+        const int64_t cap = out.numel() / std::max<int64_t>(v.row_elems, 1);
         ddstore::gather_csr(stream(), (const void* const*)v.d_peers, v.d_prefix,
                             v.d_elem_prefix, nparts_, v.d_goff,
                             idx.data_ptr<int64_t>(), nidx,
                             out_off.data_ptr<int64_t>(),
-                            v.row_elems * v.itemsize, total_elems,
+                            v.row_elems * v.itemsize, cap,
                             out.data_ptr(), v.d_oob);
         v.n_gather += 1;
         v.rows_gathered += nidx;
         v.bytes_gathered += total_elems * v.row_elems * v.itemsize;
     }
 
-    // One-call CSR fetch: lens kernel -> cumsum -> gather, no host round
-    // trips between stages. `out` is a capacity buffer (>= worst-case batch
-    // elements); returns the [n+1] element-offset tensor (device).
+    // One-call CSR fetch: ONE fused kernel (per-sample lens + decoupled-
+    // lookback exclusive scan + payload gather), no host round trips and no
+    // intermediate lens/zeros launches. `out` is a capacity buffer
+    // (>= worst-case batch elements); returns the [n+1] element-offset
+    // tensor (device). True gathered-element stats accumulate in the
+    // device counter block (the capacity would over-report).
     at::Tensor gather_csr_fast(const std::string& name, const at::Tensor& idx,
                                at::Tensor out) {
         RoctxRange rr_("ddstore::gather_csr_fast");
@@ -419,27 +499,27 @@ public:
         TORCH_CHECK(out.is_contiguous() && out.device().is_cuda() &&
                         dds_type_of(out) == v.dds_t,
                     "ddstore gather_csr: bad output tensor");
-        // NB: a hand-fused 3-kernel lens+scan was tried and measured WORSE
-        // (plan_a with 16 serial samples/thread is latency-bound at 18.8 us
-        // vs 6.4 us for the fully-parallel lens kernel + ~11 us rocprim
-        // scan); the simple pipeline below is the measured optimum.
         auto opts = at::TensorOptions().dtype(at::kLong).device(idx.device());
-        at::Tensor off = at::zeros({nidx + 1}, opts);
-        at::Tensor lens = at::empty({nidx}, opts);
-        ddstore::csr_lens(stream(), v.d_goff, idx.data_ptr<int64_t>(), nidx,
-                          v.prefix[nparts_], lens.data_ptr<int64_t>(), v.d_oob);
-        at::Tensor off_tail = off.slice(0, 1, nidx + 1);
-        at::cumsum_out(off_tail, lens, 0);
+        at::Tensor off = at::empty({nidx + 1}, opts);
+        if (nidx == 0) {
+            off.zero_();
+            return off;
+        }
         const int64_t cap = out.numel() / std::max<int64_t>(v.row_elems, 1);
-        ddstore::gather_csr(stream(), (const void* const*)v.d_peers, v.d_prefix,
-                            v.d_elem_prefix, nparts_, v.d_goff,
-                            idx.data_ptr<int64_t>(), nidx,
-                            off.data_ptr<int64_t>(),
-                            v.row_elems * v.itemsize, cap,
-                            out.data_ptr(), v.d_oob);
+        const size_t scratch = ddstore::csr_fused_scratch_bytes(nidx);
+        at::Tensor tiles = at::empty(
+            {(int64_t)scratch},
+            at::TensorOptions().dtype(at::kByte).device(idx.device()));
+        HIP_CHECK(hipMemsetAsync(tiles.data_ptr(), 0, scratch, stream()));
+        ddstore::gather_csr_fused(stream(), (const void* const*)v.d_peers,
+                                  v.d_prefix, v.d_elem_prefix, nparts_, v.d_goff,
+                                  idx.data_ptr<int64_t>(), nidx,
+                                  off.data_ptr<int64_t>(),
+                                  v.row_elems * v.itemsize, cap,
+                                  out.data_ptr(), v.d_oob, tiles.data_ptr());
         v.n_gather += 1;
         v.rows_gathered += nidx;
-        v.bytes_gathered += cap * v.row_elems * v.itemsize;
+        // bytes accounted via the device DDS_CTR_ELEMS counter (see query)
         return off;
     }
 
@@ -452,7 +532,8 @@ public:
                         lens.device().is_cuda() && lens.numel() == idx.numel(),
                     "ddstore csr_lens: bad lens tensor");
         ddstore::csr_lens(stream(), v.d_goff, idx.data_ptr<int64_t>(), idx.numel(),
-                          v.prefix[nparts_], lens.data_ptr<int64_t>(), v.d_oob);
+                          v.prefix[nparts_], lens.data_ptr<int64_t>(), v.d_oob,
+                          nullptr);
     }
 
     void scatter_local(const std::string& name, const at::Tensor& local_idx,
@@ -487,8 +568,29 @@ public:
     void epoch_end() {
         fsm_.end();
         HIP_CHECK(hipStreamSynchronize(stream()));
+        if (strict_mode()) check_strict();
     }
     bool epoch_active() const { return fsm_.fence_active; }
+
+    // Raise if any variable has skipped samples (cumulative counters; reset
+    // via reset_counters after an intentional OOB probe).
+    void check_strict() {
+        for (auto& kv : vars_) {
+            DeviceVar& v = kv.second;
+            if (!v.d_oob) continue;
+            unsigned long long ctrs[DDS_NCTR] = {0, 0, 0};
+            HIP_CHECK(hipMemcpy(ctrs, v.d_oob, DDS_NCTR * 8,
+                                hipMemcpyDeviceToHost));
+            TORCH_CHECK(ctrs[DDS_CTR_OOB] == 0, "ddstore strict: variable '",
+                        kv.first, "': ", ctrs[DDS_CTR_OOB],
+                        " out-of-range sample indices were skipped "
+                        "(DDSTORE_STRICT=1)");
+            TORCH_CHECK(ctrs[DDS_CTR_CAP] == 0, "ddstore strict: variable '",
+                        kv.first, "': ", ctrs[DDS_CTR_CAP],
+                        " CSR samples skipped: output capacity buffer too "
+                        "small (DDSTORE_STRICT=1)");
+        }
+    }
 
     py::dict query(const std::string& name) {
         DeviceVar& v = var(name);
@@ -505,14 +607,31 @@ public:
         }
         d["n_gather"] = v.n_gather;
         d["rows_gathered"] = v.rows_gathered;
-        d["bytes_gathered"] = v.bytes_gathered;
+        int64_t bytes = v.bytes_gathered;
         if (v.d_oob) {
-            unsigned long long oob = 0;
+            unsigned long long ctrs[DDS_NCTR] = {0, 0, 0};
             HIP_CHECK(hipStreamSynchronize(stream()));
-            HIP_CHECK(hipMemcpy(&oob, v.d_oob, 8, hipMemcpyDeviceToHost));
-            d["oob_skipped"] = (int64_t)oob;
+            HIP_CHECK(hipMemcpy(ctrs, v.d_oob, DDS_NCTR * 8,
+                                hipMemcpyDeviceToHost));
+            d["oob_skipped"] = (int64_t)ctrs[DDS_CTR_OOB];
+            d["cap_skipped"] = (int64_t)ctrs[DDS_CTR_CAP];
+            // gather_csr_fast accounts true elements device-side (the
+            // capacity-buffer size would over-report)
+            bytes += (int64_t)ctrs[DDS_CTR_ELEMS] * v.row_elems * v.itemsize;
         }
+        d["bytes_gathered"] = bytes;
         return d;
+    }
+
+    // Zero a variable's skip/stat counters (after an intentional OOB test,
+    // or between bench phases).
+    void reset_counters(const std::string& name) {
+        DeviceVar& v = var(name);
+        if (v.d_oob) {
+            HIP_CHECK(hipStreamSynchronize(stream()));
+            HIP_CHECK(hipMemset(v.d_oob, 0, DDS_NCTR * 8));
+        }
+        v.n_gather = v.rows_gathered = v.bytes_gathered = 0;
     }
 
     bool has(const std::string& name) const { return vars_.count(name) > 0; }
@@ -619,6 +738,7 @@ struct HostVar {
     std::vector<std::string> peer_names;
     at::Tensor goff;  // CPU int64 [ntotal+1], replicated
     int64_t n_gather = 0, rows_gathered = 0, bytes_gathered = 0;
+    int64_t oob_skipped = 0, cap_skipped = 0;
 };
 
 class HostStore {
@@ -706,6 +826,52 @@ public:
         return n;
     }
 
+    std::string init_csr(const std::string& name, int64_t nsamples,
+                         int64_t nelems, int64_t row_elems, at::ScalarType st,
+                         std::vector<int64_t> nsamples_all,
+                         std::vector<int64_t> nelems_all,
+                         const at::Tensor& goff_cpu) {
+        check_new(name);
+        at::Tensor proto = at::empty({0}, at::TensorOptions().dtype(st));
+        HostVar v;
+        v.is_csr = true;
+        v.st = st;
+        v.dds_t = dds_type_of(proto);
+        v.itemsize = dds_itemsize(v.dds_t);
+        v.row_elems = row_elems;
+        v.nrows_local = nsamples;
+        v.nelems_local = nelems;
+        v.prefix = make_prefix(nsamples_all);
+        v.elem_prefix = make_prefix(nelems_all);
+        TORCH_CHECK(goff_cpu.scalar_type() == at::kLong && goff_cpu.is_contiguous() &&
+                        goff_cpu.numel() == v.prefix[nparts_] + 1,
+                    "ddstore init_csr: bad global offsets");
+        v.goff = goff_cpu;
+        create_shm(v, name, (size_t)(nelems * row_elems * v.itemsize));
+        std::memset(v.base, 0, v.base_bytes);
+        v.active = true;
+        std::string n = v.shm_name;
+        vars_[name] = std::move(v);
+        return n;
+    }
+
+    void update_elems(const std::string& name, const at::Tensor& src,
+                      int64_t elem_offset) {
+        HostVar& v = var(name);
+        TORCH_CHECK(v.is_csr, "ddstore update_elems: CSR variables only");
+        TORCH_CHECK(src.is_contiguous() && src.device().is_cpu(),
+                    "ddstore update: array must be C-contiguous on CPU");
+        TORCH_CHECK(dds_itemsize(dds_type_of(src)) == v.itemsize,
+                    "ddstore update: itemsize mismatch");
+        int64_t nelems = src.numel() / (v.row_elems > 0 ? v.row_elems : 1);
+        TORCH_CHECK(src.numel() == nelems * v.row_elems,
+                    "ddstore update: shape mismatch");
+        TORCH_CHECK(elem_offset >= 0 && elem_offset + nelems <= v.nelems_local,
+                    "ddstore update: out of range");
+        std::memcpy((char*)v.base + elem_offset * v.row_elems * v.itemsize,
+                    src.data_ptr(), (size_t)src.numel() * v.itemsize);
+    }
+
     void open_peers(const std::string& name, const std::vector<std::string>& names) {
         HostVar& v = var(name);
         TORCH_CHECK((int)names.size() == nparts_, "ddstore open_peers: size mismatch");
@@ -778,9 +944,13 @@ public:
                         idx.device().is_cpu(),
                     "ddstore gather: indices must be contiguous int64 on CPU");
         int64_t nidx = idx.numel();
+        // dtype (not just itemsize) must match: a same-size different dtype
+        // (bf16 out from an f16 store) would silently bit-reinterpret here
+        // while the GPU path converts numerically; the Python
+        // layer converts via a staging tensor when dtypes differ.
         TORCH_CHECK(out.is_contiguous() && out.device().is_cpu() &&
-                        dds_itemsize(dds_type_of(out)) == v.itemsize,
-                    "ddstore gather: output must be contiguous CPU tensor of the store itemsize");
+                        dds_type_of(out) == v.dds_t,
+                    "ddstore gather: output must be a contiguous CPU tensor of the store dtype");
         TORCH_CHECK(out.numel() == nidx * v.row_elems, "ddstore gather: shape mismatch");
         const int64_t* ip = idx.data_ptr<int64_t>();
         const int64_t rb = v.row_elems * v.itemsize;
@@ -819,20 +989,42 @@ public:
         const int64_t* oo = out_off.data_ptr<int64_t>();
         const int64_t* go = v.goff.data_ptr<int64_t>();
         const int64_t eb = v.row_elems * v.itemsize;
+        const int64_t ntotal = v.prefix[nparts_];
+        // real buffer capacity in elements: a sample whose slice would end
+        // past it (or an out-of-range id) is skipped + counted, mirroring
+        // the device kernels -- never read/written out of bounds
+        const int64_t cap =
+            out.numel() / (v.row_elems > 0 ? v.row_elems : (int64_t)1);
         char* op = (char*)out.data_ptr();
+        std::atomic<int64_t> oob(0), capskip(0), elems(0);
         at::parallel_for(0, idx.numel(), 64, [&](int64_t b, int64_t e) {
+            int64_t my_oob = 0, my_cap = 0, my_elems = 0;
             for (int64_t s = b; s < e; ++s) {
                 int64_t g = ip[s];
+                if (g < 0 || g >= ntotal) {
+                    ++my_oob;
+                    continue;
+                }
                 int p = owner_of_host(v.prefix, g);
                 int64_t e0 = go[g], n = go[g + 1] - go[g];
+                if (oo[s] < 0 || oo[s] + n > cap) {
+                    ++my_cap;
+                    continue;
+                }
                 std::memcpy(op + oo[s] * eb,
                             (const char*)v.peers[p] + (e0 - v.elem_prefix[p]) * eb,
                             (size_t)(n * eb));
+                my_elems += n;
             }
+            oob += my_oob;
+            capskip += my_cap;
+            elems += my_elems;
         });
+        v.oob_skipped += oob.load();
+        v.cap_skipped += capskip.load();
         v.n_gather += 1;
         v.rows_gathered += idx.numel();
-        v.bytes_gathered += total_elems * eb;
+        v.bytes_gathered += elems.load() * eb;  // true gathered bytes
     }
 
     void scatter_local(const std::string& name, const at::Tensor& local_idx,
@@ -864,8 +1056,31 @@ public:
     }
 
     void epoch_begin() { fsm_.begin(); }
-    void epoch_end() { fsm_.end(); }
+    void epoch_end() {
+        fsm_.end();
+        if (strict_mode()) check_strict();
+    }
     bool epoch_active() const { return fsm_.fence_active; }
+
+    void check_strict() {
+        for (auto& kv : vars_) {
+            HostVar& v = kv.second;
+            TORCH_CHECK(v.oob_skipped == 0, "ddstore strict: variable '",
+                        kv.first, "': ", v.oob_skipped,
+                        " out-of-range sample indices were skipped "
+                        "(DDSTORE_STRICT=1)");
+            TORCH_CHECK(v.cap_skipped == 0, "ddstore strict: variable '",
+                        kv.first, "': ", v.cap_skipped,
+                        " CSR samples skipped: output capacity buffer too "
+                        "small (DDSTORE_STRICT=1)");
+        }
+    }
+
+    void reset_counters(const std::string& name) {
+        HostVar& v = var(name);
+        v.oob_skipped = v.cap_skipped = 0;
+        v.n_gather = v.rows_gathered = v.bytes_gathered = 0;
+    }
 
     py::dict query(const std::string& name) {
         HostVar& v = var(name);
@@ -883,6 +1098,8 @@ public:
         d["n_gather"] = v.n_gather;
         d["rows_gathered"] = v.rows_gathered;
         d["bytes_gathered"] = v.bytes_gathered;
+        d["oob_skipped"] = v.oob_skipped;
+        d["cap_skipped"] = v.cap_skipped;
         return d;
     }
 

@@ -39,9 +39,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("add", &DeviceStore::add)
         .def("init", &DeviceStore::init)
         .def("add_csr", &DeviceStore::add_csr)
+        .def("init_csr", &DeviceStore::init_csr)
         .def("ipc_handle", &DeviceStore::ipc_handle)
         .def("open_peers", &DeviceStore::open_peers)
         .def("update", &DeviceStore::update)
+        .def("update_elems", &DeviceStore::update_elems)
+        .def("reset_counters", &DeviceStore::reset_counters)
+        .def("check_strict", &DeviceStore::check_strict)
         .def("get_range", &DeviceStore::get_range)
         .def("gather", &DeviceStore::gather)
         .def("gather_affine", &DeviceStore::gather_affine)
@@ -65,8 +69,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("add", &HostStore::add)
         .def("init", &HostStore::init)
         .def("add_csr", &HostStore::add_csr)
+        .def("init_csr", &HostStore::init_csr)
         .def("open_peers", &HostStore::open_peers)
         .def("update", &HostStore::update)
+        .def("update_elems", &HostStore::update_elems)
+        .def("reset_counters", &HostStore::reset_counters)
+        .def("check_strict", &HostStore::check_strict)
         .def("get_range", &HostStore::get_range)
         .def("gather", &HostStore::gather)
         .def("gather_csr", &HostStore::gather_csr)

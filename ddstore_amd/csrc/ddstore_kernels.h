@@ -29,6 +29,17 @@ enum DDSType : int {
 // 8 GPUs is the design point, 128 leaves headroom for oversubscribed tests).
 #define DDS_MAX_PARTS 128
 
+// Per-variable device counter block layout (unsigned long long[DDS_NCTR]):
+//   [0] oob_skipped  -- out-of-range sample indices skipped by a gather
+//   [1] cap_skipped  -- CSR samples skipped because the output capacity
+//                       buffer was too small (never write past it)
+//   [2] csr_elems    -- true elements gathered by gather_csr_fast (stats;
+//                       the capacity-buffer size would over-report)
+#define DDS_CTR_OOB 0
+#define DDS_CTR_CAP 1
+#define DDS_CTR_ELEMS 2
+#define DDS_NCTR 3
+
 namespace ddstore {
 
 // Gather `nidx` fixed-stride rows (row = `row_elems` elements of dtype
@@ -59,7 +70,9 @@ void gather_rows_affine(hipStream_t stream,
 // CSR (variable-length record) gather: sample `g` owns elements
 // [d_goff[g], d_goff[g+1]) of the global element space; each element is
 // `elem_bytes` bytes (= disp * itemsize). Output element offsets per sample
-// are provided in d_out_off (exclusive scan, [nidx+1]).
+// are provided in d_out_off (exclusive scan, [nidx+1]). `cap_elems` is the
+// output buffer's capacity in elements: a sample whose slice would end past
+// it is SKIPPED and counted in d_ctrs[DDS_CTR_CAP] (never written OOB).
 //   d_sample_prefix : array[nparts+1] -- per-rank sample-count prefix sums
 //   d_elem_prefix   : array[nparts+1] -- per-rank element-count prefix sums
 void gather_csr(hipStream_t stream,
@@ -69,13 +82,32 @@ void gather_csr(hipStream_t stream,
                 const int64_t* d_goff,
                 const int64_t* d_idx, int64_t nidx,
                 const int64_t* d_out_off,
-                int64_t elem_bytes, int64_t total_elems,
-                void* d_out, unsigned long long* d_oob);
+                int64_t elem_bytes, int64_t cap_elems,
+                void* d_out, unsigned long long* d_ctrs);
 
-// lens[i] = goff[idx[i]+1] - goff[idx[i]] (CSR gather plan helper).
+// lens[i] = goff[idx[i]+1] - goff[idx[i]] (CSR gather plan helper). If
+// d_elems != nullptr the true total (sum of lens) is accumulated into it
+// (wave-reduced; used by gather_csr_fast for exact byte stats).
 void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
               int64_t nidx, int64_t nsamples, int64_t* d_lens,
-              unsigned long long* d_oob);
+              unsigned long long* d_oob, unsigned long long* d_elems);
+
+// Fused CSR fetch: ONE kernel computes per-sample lengths, their exclusive
+// scan (decoupled lookback across workgroups), writes d_out_off[nidx+1] and
+// gathers every sample's payload -- replaces the lens+cumsum+gather pipeline
+// (~25 us of plan overhead at B=262144 in round 1). `d_tiles` is a zeroed
+// scratch block of csr_fused_scratch_bytes(nidx); d_ctrs as in gather_csr.
+size_t csr_fused_scratch_bytes(int64_t nidx);
+void gather_csr_fused(hipStream_t stream,
+                      const void* const* d_peer_base,
+                      const int64_t* d_sample_prefix,
+                      const int64_t* d_elem_prefix, int nparts,
+                      const int64_t* d_goff,
+                      const int64_t* d_idx, int64_t nidx,
+                      int64_t* d_out_off,
+                      int64_t elem_bytes, int64_t cap_elems,
+                      void* d_out, unsigned long long* d_ctrs,
+                      void* d_tiles);
 
 
 // Scatter rows of a packed buffer into the local shard at arbitrary local row

@@ -284,7 +284,8 @@ k_gather_csr_wave(const void* const* peer_base,
                   const int64_t* goff,
                   const int64_t* idx, int64_t nidx,
                   const int64_t* out_off, int64_t chunks_per_elem,
-                  T* __restrict__ out, unsigned long long* oob) {
+                  int64_t cap_elems,
+                  T* __restrict__ out, unsigned long long* ctrs) {
     __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
     __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
     __shared__ const T* s_base[DDS_MAX_PARTS];
@@ -303,14 +304,20 @@ k_gather_csr_wave(const void* const* peer_base,
     for (int64_t s = first; s < nidx; s += step) {
         const int64_t g = idx[s];
         if (g < 0 || g >= s_sprefix[nparts]) {
-            if (tid == 0) atomicAdd(oob, 1ull);
+            if (tid == 0) atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
             continue;
         }
         const int p = owner_of(s_sprefix, nparts, g);
         const int64_t e0 = goff[g];
-        const int64_t nch = (goff[g + 1] - e0) * chunks_per_elem;
+        const int64_t len = goff[g + 1] - e0;
+        const int64_t o0 = out_off[s];
+        if (o0 < 0 || o0 + len > cap_elems) {  // undersized capacity buffer:
+            if (tid == 0) atomicAdd(ctrs + DDS_CTR_CAP, 1ull);  // never write OOB
+            continue;
+        }
+        const int64_t nch = len * chunks_per_elem;
         const T* src = s_base[p] + (e0 - s_eprefix[p]) * chunks_per_elem;
-        T* dst = out + out_off[s] * chunks_per_elem;
+        T* dst = out + o0 * chunks_per_elem;
         for (int64_t c = tid; c < nch; c += GROUP) dst[c] = src[c];
     }
 }
@@ -320,7 +327,9 @@ k_gather_csr_wave(const void* const* peer_base,
 // the equivalent `goff[idx+1]-goff[idx]` costs per step).
 __global__ void __launch_bounds__(kBlock)
 k_csr_lens(const int64_t* goff, const int64_t* idx, int64_t nidx, int64_t nsamples,
-           int64_t* __restrict__ lens, unsigned long long* oob) {
+           int64_t* __restrict__ lens, unsigned long long* oob,
+           unsigned long long* elems) {
+    int64_t acc = 0;
     for (int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x; i < nidx;
          i += (int64_t)gridDim.x * kBlock) {
         const int64_t g = idx[i];
@@ -329,7 +338,16 @@ k_csr_lens(const int64_t* goff, const int64_t* idx, int64_t nidx, int64_t nsampl
             atomicAdd(oob, 1ull);
             continue;
         }
-        lens[i] = goff[g + 1] - goff[g];
+        const int64_t L = goff[g + 1] - goff[g];
+        lens[i] = L;
+        acc += L;
+    }
+    if (elems) {
+        // wave-reduce (64-wide) then one atomic per wave: exact gathered-
+        // element stats without per-thread atomic contention
+        for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+        if ((threadIdx.x & 63) == 0 && acc)
+            atomicAdd(elems, (unsigned long long)acc);
     }
 }
 
@@ -368,7 +386,8 @@ k_gather_csr_dw(const void* const* peer_base,
                 const int64_t* goff,
                 const int64_t* idx, int64_t nidx,
                 const int64_t* out_off, int64_t dwords_per_elem,
-                uint32_t* __restrict__ out, unsigned long long* oob) {
+                int64_t cap_elems,
+                uint32_t* __restrict__ out, unsigned long long* ctrs) {
     __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
     __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
     __shared__ const uint32_t* s_base[DDS_MAX_PARTS];
@@ -387,15 +406,151 @@ k_gather_csr_dw(const void* const* peer_base,
     for (int64_t s = first; s < nidx; s += step) {
         const int64_t g = idx[s];
         if (g < 0 || g >= s_sprefix[nparts]) {
-            if (tid == 0) atomicAdd(oob, 1ull);
+            if (tid == 0) atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
             continue;
         }
         const int p = owner_of(s_sprefix, nparts, g);
         const int64_t e0 = goff[g];
-        const int64_t nd = (goff[g + 1] - e0) * dwords_per_elem;
-        copy_dwords_store16(out + out_off[s] * dwords_per_elem,
+        const int64_t len = goff[g + 1] - e0;
+        const int64_t o0 = out_off[s];
+        if (o0 < 0 || o0 + len > cap_elems) {
+            if (tid == 0) atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
+            continue;
+        }
+        copy_dwords_store16(out + o0 * dwords_per_elem,
                             s_base[p] + (e0 - s_eprefix[p]) * dwords_per_elem,
-                            nd, tid, GROUP);
+                            len * dwords_per_elem, tid, GROUP);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fused CSR fetch: per-sample lens + exclusive scan + gather in ONE kernel.
+//
+// Round-1 measured the separate plan (zeros + lens kernel + torch cumsum) at
+// ~25 us/step (B=262144); a naive serial-lens fusion was tried and reverted
+// (latency-bound). This version keeps the scan fully parallel with a
+// decoupled-lookback across workgroup tiles of kBlock samples: each tile
+// block-scans its 256 lens, publishes its aggregate to `tile_state`,
+// looks back over earlier tiles for its exclusive base (values arrive long
+// before payload copies finish, so the chain pipelines under the gather),
+// then gathers its samples' payloads with GROUP lanes per sample.
+//
+// tile_state[t] encodes (value << 2) | flag, flag 1 = aggregate ready,
+// 2 = inclusive prefix ready; it must be zeroed before launch. The grid is
+// capped to the occupancy-resident block count so every spinning tile's
+// producer is guaranteed to be scheduled (forward progress).
+// ---------------------------------------------------------------------------
+template <typename T, int GROUP, bool DW16>
+__global__ void __launch_bounds__(kBlock)
+k_gather_csr_fused(const void* const* peer_base,
+                   const int64_t* sample_prefix, const int64_t* elem_prefix,
+                   int nparts, const int64_t* goff,
+                   const int64_t* idx, int64_t nidx,
+                   int64_t* __restrict__ out_off, int64_t chunks_per_elem,
+                   int64_t cap_elems, T* __restrict__ out,
+                   unsigned long long* ctrs,
+                   unsigned long long* __restrict__ tile_state) {
+    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
+    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
+    __shared__ const T* s_base[DDS_MAX_PARTS];
+    __shared__ int64_t s_g[kBlock];     // global sample id (-1 = oob)
+    __shared__ int64_t s_len[kBlock];   // sample length (elements)
+    __shared__ int64_t s_excl[kBlock];  // global exclusive output offset
+    __shared__ int64_t s_wsum[kBlock / 64];
+    __shared__ int64_t s_tile_base;
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
+        s_sprefix[i] = sample_prefix[i];
+        s_eprefix[i] = elem_prefix[i];
+    }
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const T*>(peer_base[i]);
+    __syncthreads();
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
+    for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+        const int64_t i = tile * kBlock + threadIdx.x;
+        int64_t g = -1, L = 0;
+        if (i < nidx) {
+            g = idx[i];
+            if (g < 0 || g >= s_sprefix[nparts]) {
+                atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
+                g = -1;
+            } else {
+                L = goff[g + 1] - goff[g];
+            }
+        }
+        s_g[threadIdx.x] = g;
+        s_len[threadIdx.x] = L;
+        // 64-wide inclusive wave scan of L, then cross-wave bases via LDS
+        int64_t x = L;
+        for (int off = 1; off < 64; off <<= 1) {
+            int64_t y = __shfl_up(x, off);
+            if (lane >= off) x += y;
+        }
+        if (lane == 63) s_wsum[wave] = x;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int64_t acc = 0;
+            for (int w = 0; w < kBlock / 64; ++w) {
+                int64_t t = s_wsum[w];
+                s_wsum[w] = acc;
+                acc += t;
+            }
+            __threadfence();
+            atomicExch(&tile_state[tile],
+                       ((unsigned long long)acc << 2) | 1ull);
+            int64_t excl = 0;
+            for (int64_t t = tile - 1; t >= 0; --t) {
+                unsigned long long v;
+                do {
+                    v = atomicAdd(&tile_state[t], 0ull);  // atomic read
+                } while ((v & 3ull) == 0ull);
+                excl += (int64_t)(v >> 2);
+                if ((v & 3ull) == 2ull) break;
+            }
+            __threadfence();
+            atomicExch(&tile_state[tile],
+                       ((unsigned long long)(excl + acc) << 2) | 2ull);
+            s_tile_base = excl;
+            atomicAdd(ctrs + DDS_CTR_ELEMS, (unsigned long long)acc);
+            if (tile == 0) out_off[0] = 0;
+        }
+        __syncthreads();
+        const int64_t my_excl = s_tile_base + s_wsum[wave] + (x - L);
+        s_excl[threadIdx.x] = my_excl;
+        if (i < nidx) out_off[i + 1] = my_excl + L;
+        __syncthreads();
+        // gather phase: GROUP lanes cooperate per sample of this tile
+        const int nsamp = (int)(nidx - tile * kBlock < kBlock
+                                    ? nidx - tile * kBlock : (int64_t)kBlock);
+        const int gid = threadIdx.x / GROUP;
+        const int tid = threadIdx.x % GROUP;
+        constexpr int GPB = kBlock / GROUP;
+        for (int s = gid; s < nsamp; s += GPB) {
+            const int64_t g2 = s_g[s];
+            if (g2 < 0) continue;
+            const int64_t len = s_len[s];
+            const int64_t o0 = s_excl[s];
+            if (o0 + len > cap_elems) {
+                if (tid == 0) atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
+                continue;
+            }
+            const int p = owner_of(s_sprefix, nparts, g2);
+            const int64_t e0 = goff[g2];
+            const T* src = s_base[p] + (e0 - s_eprefix[p]) * chunks_per_elem;
+            T* dst = out + o0 * chunks_per_elem;
+            if constexpr (DW16) {
+                copy_dwords_store16(reinterpret_cast<uint32_t*>(dst),
+                                    reinterpret_cast<const uint32_t*>(src),
+                                    len * chunks_per_elem, tid, GROUP);
+            } else {
+                const int64_t nch = len * chunks_per_elem;
+                for (int64_t c = tid; c < nch; c += GROUP) dst[c] = src[c];
+            }
+        }
+        __syncthreads();  // LDS tiles reused next iteration
     }
 }
 
@@ -459,7 +614,7 @@ void launch_gather_cast_one(hipStream_t stream, const void* const* pb,
                             int64_t n, int64_t re, Tout* out,
                             unsigned long long* oob) {
     constexpr int VEC = 16 / sizeof(Tout);
-    if (re % VEC == 0) {
+    if (re % VEC == 0 && ((uintptr_t)out % 16) == 0) {
         const int grid = n_blocks(n * (re / VEC));
         hipLaunchKernelGGL((k_gather_rows_castv<Tin, Tout>), dim3(grid), dim3(kBlock),
                            0, stream, pb, pf, np, idx, n, re, out, oob);
@@ -503,7 +658,7 @@ void launch_affine_out(hipStream_t stream, const void* const* pb, const int64_t*
 #define DDS_AFF(tag, T)                                                              \
     case tag: {                                                                      \
         constexpr int VEC = 16 / sizeof(T);                                          \
-        if (re % VEC == 0) {                                                         \
+        if (re % VEC == 0 && ((uintptr_t)out % 16) == 0) {                           \
             const int grid = n_blocks(n * (re / VEC));                               \
             hipLaunchKernelGGL((k_gather_rows_affine<Tin, T>), dim3(grid),           \
                                dim3(kBlock), 0, stream, pb, pf, np, idx, n, re, a,   \
@@ -560,7 +715,11 @@ void gather_rows(hipStream_t stream,
                  void* d_out, unsigned long long* d_oob) {
     if (nidx == 0 || row_elems == 0) return;
     const int64_t row_bytes = row_elems * dds_itemsize(in_t);
-    if (in_t == out_t && row_bytes % 16 == 0) {
+    // the uint4/VecT paths require a 16/32-B-aligned output pointer; an
+    // offset view (e.g. buf[1:]) falls through to the cast/scalar kernels,
+    // whose own vector path re-checks alignment
+    const uintptr_t out_align = (uintptr_t)d_out;
+    if (in_t == out_t && row_bytes % 16 == 0 && out_align % 16 == 0) {
         const int64_t cpr = row_bytes / 16;
         // NB: a thread-per-row variant (one idx load, all chunks in flight
         // per thread) was measured SLOWER on MI355X for <=64 B rows (7.5 vs
@@ -571,7 +730,7 @@ void gather_rows(hipStream_t stream,
         // A/B on MI355X (B=131072): 512 B rows -- 16B 30.1us / 32B 28.5 /
         // 64B 33.9; 64 B rows -- 16B 6.1 / 32B 6.7 (too few threads per
         // row). 32-B chunks win once a row has >=4 of them.
-        if (row_bytes % 32 == 0 && row_bytes >= 128) {
+        if (row_bytes % 32 == 0 && row_bytes >= 128 && out_align % 32 == 0) {
             const int64_t cpr32 = row_bytes / 32;
             const int grid = n_blocks(nidx * cpr32);
             hipLaunchKernelGGL((k_gather_rows_b16<32>), dim3(grid), dim3(kBlock), 0,
@@ -606,10 +765,10 @@ void gather_rows(hipStream_t stream,
 
 void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
               int64_t nidx, int64_t nsamples, int64_t* d_lens,
-              unsigned long long* d_oob) {
+              unsigned long long* d_oob, unsigned long long* d_elems) {
     if (nidx == 0) return;
     hipLaunchKernelGGL(k_csr_lens, dim3(n_blocks(nidx)), dim3(kBlock), 0, stream,
-                       d_goff, d_idx, nidx, nsamples, d_lens, d_oob);
+                       d_goff, d_idx, nidx, nsamples, d_lens, d_oob, d_elems);
 }
 
 
@@ -620,8 +779,8 @@ void gather_csr(hipStream_t stream,
                 const int64_t* d_goff,
                 const int64_t* d_idx, int64_t nidx,
                 const int64_t* d_out_off,
-                int64_t elem_bytes, int64_t total_elems,
-                void* d_out, unsigned long long* d_oob) {
+                int64_t elem_bytes, int64_t cap_elems,
+                void* d_out, unsigned long long* d_ctrs) {
     if (nidx == 0) return;
     // lanes cooperating per sample: small groups amortize the per-sample
     // setup (row-id load + directory search) across the wave and keep more
@@ -629,7 +788,6 @@ void gather_csr(hipStream_t stream,
     // 32 KiB average samples (2.23 vs 1.44 G samples/s at 512 B vs GROUP=64)
     // -- the only reason to grow the group is chip COVERAGE when the batch
     // has few samples (need >=1024 workgroups across 256 CUs).
-    (void)total_elems;
     static const int g_override = [] {
         const char* e = getenv("DDSTORE_CSR_GROUP");
         return e ? atoi(e) : 0;
@@ -638,11 +796,14 @@ void gather_csr(hipStream_t stream,
     if (g_override) group = g_override;
     int64_t b = (nidx + (kBlock / group) - 1) / (kBlock / group);
     const int grid = (int)(b < kMaxBlocks ? b : kMaxBlocks);
+    // vector paths reinterpret d_out; a misaligned output (e.g. an offset
+    // view) falls back to the next-narrower granularity
+    const uintptr_t oa = (uintptr_t)d_out;
 #define DDS_CSR_G(T, div, G)                                                         \
     hipLaunchKernelGGL((k_gather_csr_wave<T, G>), dim3(grid), dim3(kBlock), 0,       \
                        stream, d_peer_base, d_sample_prefix, d_elem_prefix, nparts,  \
-                       d_goff, d_idx, nidx, d_out_off, elem_bytes / div, (T*)d_out,  \
-                       d_oob)
+                       d_goff, d_idx, nidx, d_out_off, elem_bytes / div, cap_elems,  \
+                       (T*)d_out, d_ctrs)
 #define DDS_CSR_T(T, div)                                                            \
     do {                                                                             \
         if (group == 8) DDS_CSR_G(T, div, 8);                                        \
@@ -653,11 +814,11 @@ void gather_csr(hipStream_t stream,
 #define DDS_CSR_DW(G)                                                                \
     hipLaunchKernelGGL((k_gather_csr_dw<G>), dim3(grid), dim3(kBlock), 0, stream,    \
                        d_peer_base, d_sample_prefix, d_elem_prefix, nparts, d_goff,  \
-                       d_idx, nidx, d_out_off, elem_bytes / 4, (uint32_t*)d_out,     \
-                       d_oob)
-    if (elem_bytes % 16 == 0) {
+                       d_idx, nidx, d_out_off, elem_bytes / 4, cap_elems,            \
+                       (uint32_t*)d_out, d_ctrs)
+    if (elem_bytes % 16 == 0 && oa % 16 == 0) {
         DDS_CSR_T(uint4, 16);
-    } else if (elem_bytes % 4 == 0) {
+    } else if (elem_bytes % 4 == 0 && oa % 4 == 0) {
         // 4/8-B-granular elements: dword addressing, stores re-aligned to
         // dwordx4 inside each sample's payload
         if (group == 8) DDS_CSR_DW(8);
@@ -671,6 +832,98 @@ void gather_csr(hipStream_t stream,
 #undef DDS_CSR_T
 #undef DDS_CSR_DW
 }
+size_t csr_fused_scratch_bytes(int64_t nidx) {
+    const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
+    return (size_t)(ntiles > 0 ? ntiles : 1) * sizeof(unsigned long long);
+}
+
+namespace {
+
+// Occupancy-resident grid cap for the fused kernel: the decoupled-lookback
+// spin requires every earlier tile's producer block to be scheduled, so the
+// grid must not exceed the number of simultaneously-resident blocks.
+template <typename T, int GROUP, bool DW16>
+int fused_resident_grid() {
+    static const int cached = [] {
+        int dev = 0;
+        (void)hipGetDevice(&dev);
+        hipDeviceProp_t prop{};
+        (void)hipGetDeviceProperties(&prop, dev);
+        int occ = 0;
+        (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &occ, reinterpret_cast<const void*>(&k_gather_csr_fused<T, GROUP, DW16>),
+            kBlock, 0);
+        if (occ < 1) occ = 1;
+        int sms = prop.multiProcessorCount > 0 ? prop.multiProcessorCount : 64;
+        return occ * sms;
+    }();
+    return cached;
+}
+
+template <typename T, int GROUP, bool DW16>
+void launch_csr_fused(hipStream_t stream, const void* const* pb,
+                      const int64_t* sp, const int64_t* ep, int np,
+                      const int64_t* goff, const int64_t* idx, int64_t nidx,
+                      int64_t* out_off, int64_t cpe, int64_t cap, void* out,
+                      unsigned long long* ctrs, unsigned long long* tiles) {
+    const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
+    int grid = fused_resident_grid<T, GROUP, DW16>();
+    if ((int64_t)grid > ntiles) grid = (int)ntiles;
+    hipLaunchKernelGGL((k_gather_csr_fused<T, GROUP, DW16>), dim3(grid),
+                       dim3(kBlock), 0, stream, pb, sp, ep, np, goff, idx, nidx,
+                       out_off, cpe, cap, (T*)out, ctrs, tiles);
+}
+
+} // namespace
+
+void gather_csr_fused(hipStream_t stream,
+                      const void* const* d_peer_base,
+                      const int64_t* d_sample_prefix,
+                      const int64_t* d_elem_prefix, int nparts,
+                      const int64_t* d_goff,
+                      const int64_t* d_idx, int64_t nidx,
+                      int64_t* d_out_off,
+                      int64_t elem_bytes, int64_t cap_elems,
+                      void* d_out, unsigned long long* d_ctrs,
+                      void* d_tiles) {
+    if (nidx == 0) return;
+    static const int g_override = [] {
+        const char* e = getenv("DDSTORE_CSR_GROUP");
+        return e ? atoi(e) : 0;
+    }();
+    int group = nidx >= 16384 ? 16 : (nidx >= 4096 ? 64 : 256);
+    if (g_override) group = g_override;
+    if (group < 16) group = 16;  // fused tiles use >=16-lane sample groups
+    const uintptr_t oa = (uintptr_t)d_out;
+    auto* tiles = reinterpret_cast<unsigned long long*>(d_tiles);
+#define DDS_CSRF(T, div, DW)                                                         \
+    do {                                                                             \
+        if (group == 16)                                                             \
+            launch_csr_fused<T, 16, DW>(stream, d_peer_base, d_sample_prefix,        \
+                                        d_elem_prefix, nparts, d_goff, d_idx, nidx,  \
+                                        d_out_off, elem_bytes / div, cap_elems,      \
+                                        d_out, d_ctrs, tiles);                       \
+        else if (group == 64)                                                        \
+            launch_csr_fused<T, 64, DW>(stream, d_peer_base, d_sample_prefix,        \
+                                        d_elem_prefix, nparts, d_goff, d_idx, nidx,  \
+                                        d_out_off, elem_bytes / div, cap_elems,      \
+                                        d_out, d_ctrs, tiles);                       \
+        else                                                                         \
+            launch_csr_fused<T, 256, DW>(stream, d_peer_base, d_sample_prefix,       \
+                                         d_elem_prefix, nparts, d_goff, d_idx, nidx, \
+                                         d_out_off, elem_bytes / div, cap_elems,     \
+                                         d_out, d_ctrs, tiles);                      \
+    } while (0)
+    if (elem_bytes % 16 == 0 && oa % 16 == 0) {
+        DDS_CSRF(uint4, 16, false);
+    } else if (elem_bytes % 4 == 0 && oa % 4 == 0) {
+        DDS_CSRF(uint32_t, 4, true);
+    } else {
+        DDS_CSRF(uint8_t, 1, false);
+    }
+#undef DDS_CSRF
+}
+
 void scatter_rows_local(hipStream_t stream,
                         void* d_base, int64_t nrows_local,
                         int64_t row_elems, int elem_t,
@@ -678,13 +931,14 @@ void scatter_rows_local(hipStream_t stream,
                         const void* d_src, unsigned long long* d_oob) {
     if (nidx == 0 || row_elems == 0) return;
     const int64_t row_bytes = row_elems * dds_itemsize(elem_t);
-    if (row_bytes % 32 == 0 && row_bytes >= 128) {
+    const uintptr_t al = (uintptr_t)d_src | (uintptr_t)d_base;
+    if (row_bytes % 32 == 0 && row_bytes >= 128 && al % 32 == 0) {
         const int64_t cpr = row_bytes / 32;
         const int grid = n_blocks(nidx * cpr);
         hipLaunchKernelGGL((k_scatter_rows_b16<32>), dim3(grid), dim3(kBlock), 0,
                            stream, (uint4*)d_base, nrows_local, cpr, d_local_idx,
                            nidx, (const uint4*)d_src, d_oob);
-    } else if (row_bytes % 16 == 0) {
+    } else if (row_bytes % 16 == 0 && al % 16 == 0) {
         const int64_t cpr = row_bytes / 16;
         const int grid = n_blocks(nidx * cpr);
         hipLaunchKernelGGL((k_scatter_rows_b16<16>), dim3(grid), dim3(kBlock), 0,

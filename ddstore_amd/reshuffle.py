@@ -24,6 +24,26 @@ import torch
 from .store import DDStore
 
 
+def _check_perm_agreement(store: DDStore, perm: torch.Tensor) -> None:
+    """Every rank derives the permutation independently from the seed
+    (device RNG); if any rank disagreed (different ROCm version/arch mixing)
+    the exchange would silently corrupt data. A strided checksum allgather
+    catches divergence before any data moves."""
+    if store.size == 1:
+        return
+    n = perm.numel()
+    stride = max(1, n // 4096)
+    sample = perm[::stride]
+    pos = torch.arange(1, sample.numel() + 1, device=sample.device, dtype=torch.int64)
+    h = int(((sample + 1) * pos).sum().item())  # position-weighted, wraps ok
+    if len(set(store.comm.allgather(h))) != 1:
+        raise RuntimeError(
+            "ddstore reshuffle: ranks derived DIFFERENT permutations from the "
+            "same seed (device RNG mismatch across processes -- mixed ROCm "
+            "versions or GPU architectures?); aborting before data exchange"
+        )
+
+
 def reshuffle_epoch(store: DDStore, name: str, seed: int) -> None:
     if store._backend.epoch_active():
         raise RuntimeError(
@@ -44,6 +64,7 @@ def reshuffle_epoch(store: DDStore, name: str, seed: int) -> None:
     g = torch.Generator(device=dev)
     g.manual_seed(int(seed))
     perm = torch.randperm(ntotal, generator=g, device=dev)
+    _check_perm_agreement(store, perm)
     # rank that currently holds the row destined for slot j
     src_of_slot = torch.searchsorted(prefix, perm, right=True) - 1
 
@@ -92,6 +113,7 @@ def _reshuffle_csr(store: DDStore, name: str, seed: int) -> None:
     g = torch.Generator(device=dev)
     g.manual_seed(int(seed))
     perm = torch.randperm(ntotal, generator=g, device=dev)
+    _check_perm_agreement(store, perm)
     mine = perm[p0:p1].contiguous()
 
     values, _ = store.get_csr(name, mine)

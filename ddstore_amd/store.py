@@ -241,34 +241,86 @@ class DDStore:
         t = _as_tensor(arr)
         self._backend.update(name, self._staged(t), int(offset))
 
+    def _csr_directory(self, lens: torch.Tensor):
+        """Collective: allgather per-rank sample/element counts and build the
+        replicated global offset directory goff[ntotal+1]."""
+        nsamples = int(lens.numel())
+        nelems = int(lens.sum().item()) if nsamples else 0
+        nsamples_all = self.comm.allgather(nsamples)
+        nelems_all = self.comm.allgather(nelems)
+        lens_all = self.comm.allgather(lens.numpy())
+        goff = np.zeros(sum(nsamples_all) + 1, dtype=np.int64)
+        np.cumsum(np.concatenate(lens_all), out=goff[1:])
+        return nsamples, nelems, nsamples_all, nelems_all, torch.from_numpy(goff)
+
+    def _register_csr_meta(self, name: str, dtype, row_elems: int,
+                           nsamples_all, goff_t: torch.Tensor) -> None:
+        meta = self._mkmeta(True, dtype, row_elems, nsamples_all)
+        meta["goff"] = goff_t
+        if self.mode == "hip":
+            meta["goff_dev"] = goff_t.to(self.device)
+        self._vars[name] = meta
+        self._exchange_and_open(name)
+
     def add_csr(self, name: str, values: ArrayLike, lengths: ArrayLike) -> None:
         """Register variable-length (CSR) samples: ``lengths[i]`` elements per
         local sample, elements of fixed feature width. First-class version of
         the reference's disp=1 element-addressed convention."""
         v = _as_tensor(values)
         lens = _as_tensor(lengths).to(torch.int64).cpu()
-        nsamples = int(lens.numel())
-        nelems = int(lens.sum().item()) if nsamples else 0
+        nsamples, nelems, nsamples_all, nelems_all, goff_t = self._csr_directory(lens)
         row_elems = v.numel() // nelems if nelems > 0 else -1
         if nelems > 0 and v.numel() != nelems * row_elems:
             raise ValueError("ddstore add_csr: values size does not match lengths")
         row_elems = self._validate_uniform(row_elems, v.dtype)
-        nsamples_all = self.comm.allgather(nsamples)
-        nelems_all = self.comm.allgather(nelems)
-        lens_all = self.comm.allgather(lens.numpy())
-        goff = np.zeros(sum(nsamples_all) + 1, dtype=np.int64)
-        np.cumsum(np.concatenate(lens_all), out=goff[1:])
-        goff_t = torch.from_numpy(goff)
         self._backend.add_csr(
             name, self._staged(v), nsamples, nelems, row_elems,
             nsamples_all, nelems_all, goff_t,
         )
-        meta = self._mkmeta(True, v.dtype, row_elems, nsamples_all)
-        meta["goff"] = goff_t
-        if self.mode == "hip":
-            meta["goff_dev"] = goff_t.to(self.device)
-        self._vars[name] = meta
-        self._exchange_and_open(name)
+        self._register_csr_meta(name, v.dtype, row_elems, nsamples_all, goff_t)
+
+    def init_csr(
+        self,
+        name: str,
+        lengths: ArrayLike,
+        disp: int = 1,
+        dtype: torch.dtype = torch.float32,
+    ) -> None:
+        """Pre-allocate a zeroed CSR variable whose per-sample LENGTHS are
+        fixed now (they define the global offset directory) and whose values
+        are filled later with :meth:`update_csr` -- the reference's
+        incremental-fill pattern (init+update, ddstore.hpp:110-195,
+        README.md:107) extended to the first-class CSR layout (collective)."""
+        if dtype not in _SUPPORTED:
+            raise TypeError(f"ddstore: unsupported dtype {dtype}")
+        lens = _as_tensor(lengths).to(torch.int64).cpu()
+        nsamples, nelems, nsamples_all, nelems_all, goff_t = self._csr_directory(lens)
+        disp = self._validate_uniform(int(disp), dtype)
+        st = torch.empty(0, dtype=dtype).dtype
+        self._backend.init_csr(
+            name, nsamples, nelems, disp, st, nsamples_all, nelems_all, goff_t
+        )
+        self._register_csr_meta(name, dtype, disp, nsamples_all, goff_t)
+
+    def update_csr(self, name: str, values: ArrayLike, offset: int = 0) -> None:
+        """Local fill of a CSR variable starting at local sample ``offset``:
+        ``values`` holds the elements of samples ``offset, offset+1, ...``
+        (any contiguous run). No communication, no epoch required -- the CSR
+        analog of :meth:`update`. Like the reference, separating updates from
+        remote reads is the caller's job."""
+        meta = self._meta(name)
+        if not meta["is_csr"]:
+            raise ValueError(f"ddstore update_csr: '{name}' is not a CSR variable")
+        v = _as_tensor(values)
+        q = self._backend.query(name)
+        p0 = int(q["prefix"][self.rank])
+        ep0 = int(q["elem_prefix"][self.rank])
+        nloc = int(q["nrows_local"])
+        offset = int(offset)
+        if not 0 <= offset <= nloc:
+            raise IndexError(f"ddstore update_csr: sample offset {offset} out of range")
+        elem_off = int(meta["goff"][p0 + offset].item()) - ep0
+        self._backend.update_elems(name, self._staged(v), elem_off)
 
     # ------------------------------------------------------------------- reads
     def get(self, name: str, out: ArrayLike, start: int = 0) -> None:
@@ -469,6 +521,14 @@ class DDStore:
     def query(self, name: str) -> dict:
         return self._backend.query(name)
 
+    def reset_counters(self, name: str) -> None:
+        """Zero a variable's skip/traffic counters (e.g. after an intentional
+        out-of-range probe under ``DDSTORE_STRICT=1``)."""
+        self._backend.reset_counters(name)
+        meta = self._vars.get(name)
+        if meta is not None and "peer_rows" in meta:
+            meta["peer_rows"].zero_()
+
     def variables(self) -> list:
         """Names of all registered variables."""
         return list(self._vars)
@@ -478,7 +538,15 @@ class DDStore:
         for name in list(self._vars):
             q = self._backend.query(name)
             out[name] = {
-                k: q[k] for k in ("n_gather", "rows_gathered", "bytes_gathered")
+                k: q[k]
+                for k in (
+                    "n_gather",
+                    "rows_gathered",
+                    "bytes_gathered",
+                    "oob_skipped",
+                    "cap_skipped",
+                )
+                if k in q
             }
             meta = self._vars[name]
             if self._stats_enabled and "peer_rows" in meta:
