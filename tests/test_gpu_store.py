@@ -409,3 +409,119 @@ def test_csr_plan_scan_matches_cumsum(store, nidx):
         g = int(idx[k])
         seg = out[off_h[k] : off_h[k + 1]].cpu()
         assert torch.equal(seg, vals[int(goff[g]) : int(goff[g + 1])])
+
+
+def test_csr_fused_large_multitile(store):
+    # many lookback tiles (600k samples = ~2344 tiles) + payload check:
+    # stresses the decoupled-lookback chain across grid-stride tile rounds
+    rng = np.random.default_rng(99)
+    lens = rng.integers(0, 9, size=20000)
+    vals = torch.arange(int(lens.sum()), dtype=torch.float32).reshape(-1, 1)
+    store.add_csr("big", vals, lens)
+    nidx = 600_000
+    idx = torch.from_numpy(rng.integers(0, 20000, size=nidx)).cuda()
+    out = torch.empty(nidx * 9, 1, device="cuda:0")
+    off = store._backend.gather_csr_fast("big", idx, out)
+    torch.cuda.synchronize()
+    goff = torch.from_numpy(np.concatenate([[0], np.cumsum(lens)])).cuda()
+    ref_off = torch.zeros(nidx + 1, dtype=torch.int64, device="cuda:0")
+    torch.cumsum(goff[idx + 1] - goff[idx], 0, out=ref_off[1:])
+    assert torch.equal(off, ref_off)
+    # spot-check payloads at tile boundaries and random positions
+    off_h = off.cpu()
+    goff_h = goff.cpu()
+    for k in [0, 255, 256, 257, 65535, 65536, nidx - 1, 300_001]:
+        g = int(idx[k])
+        seg = out[int(off_h[k]) : int(off_h[k + 1])].cpu()
+        assert torch.equal(seg, vals[int(goff_h[g]) : int(goff_h[g + 1])]), k
+
+
+def test_csr_capacity_clamp_gpu(store):
+    lens = [4, 4, 4, 4]
+    vals = torch.arange(16, dtype=torch.float32).reshape(16, 1)
+    store.add_csr("clamp", vals, lens)
+    guard = torch.full((64,), -7.0, device="cuda:0")  # heap canary after out
+    out = torch.zeros(8, 1, device="cuda:0")
+    v, off = store.get_csr("clamp", [0, 1, 2, 3], out=out)
+    torch.cuda.synchronize()
+    q = store.query("clamp")
+    assert q["cap_skipped"] == 2, q
+    assert torch.equal(out.cpu(), vals[:8])
+    assert (guard == -7.0).all()  # nothing wrote past the capacity buffer
+    assert q["bytes_gathered"] == 8 * 4  # true bytes, not 4-sample total
+
+
+def test_csr_fused_true_byte_stats(store):
+    lens = [3, 1]
+    vals = torch.arange(4, dtype=torch.float32).reshape(4, 1)
+    store.add_csr("tb", vals, lens)
+    big = torch.zeros(4096, 1, device="cuda:0")  # heavily oversized capacity
+    store.get_csr("tb", [0, 1], out=big)
+    torch.cuda.synchronize()
+    assert store.query("tb")["bytes_gathered"] == 4 * 4
+
+
+def test_misaligned_output_view_gpu(store):
+    # an offset view of a buffer is contiguous but not 16-B aligned: the
+    # dispatch must fall back to scalar kernels, not emit misaligned vector
+    # stores
+    arr = torch.randn(64, 8)
+    store.add("mis", arr)
+    back = torch.empty(65 * 8 + 1, device="cuda:0")
+    out = back[1 : 1 + 64 * 8].view(64, 8)  # 4-B-aligned only
+    idx = torch.arange(64, device="cuda:0")
+    store.gather_into("mis", idx, out)
+    torch.cuda.synchronize()
+    assert torch.equal(out.cpu(), arr)
+    # same-dtype byte-move path with a bf16 store (2-B elements)
+    arr16 = torch.randn(32, 16, dtype=torch.bfloat16)
+    store.add("mis16", arr16)
+    back16 = torch.empty(33 * 16, dtype=torch.bfloat16, device="cuda:0")
+    out16 = back16[1 : 1 + 32 * 16].view(32, 16)  # 2-B-aligned only
+    store.gather_into("mis16", torch.arange(32, device="cuda:0"), out16)
+    torch.cuda.synchronize()
+    assert torch.equal(out16.cpu(), arr16)
+
+
+def test_csr_init_update_gpu(store):
+    lens = torch.tensor([3, 1, 4, 2])
+    store.init_csr("icu", lens, disp=2, dtype=torch.float32)
+    v, off = store.get_csr("icu", [0, 1, 2, 3])
+    torch.cuda.synchronize()
+    assert v.abs().sum() == 0 and off.cpu().tolist() == [0, 3, 4, 8, 10]
+    part0 = torch.arange(8, dtype=torch.float32).reshape(4, 2)
+    part1 = 100 + torch.arange(12, dtype=torch.float32).reshape(6, 2)
+    store.update_csr("icu", part0, offset=0)
+    store.update_csr("icu", part1, offset=2)
+    v, _ = store.get_csr("icu", [0, 1, 2, 3])
+    torch.cuda.synchronize()
+    assert torch.equal(v.cpu(), torch.cat([part0, part1]))
+
+
+def test_strict_mode_gpu_subprocess():
+    import os
+    import subprocess
+    import sys
+
+    code = """
+import torch
+from ddstore_amd import DDStore
+s = DDStore(device="cuda:0")
+s.add("x", torch.randn(8, 8))
+s.epoch_begin()
+out = s.get_batch("x", [0, 99])  # 99 out of range
+torch.cuda.synchronize()
+try:
+    s.epoch_end()
+    print("NO-RAISE")
+except RuntimeError as e:
+    assert "DDSTORE_STRICT" in str(e), e
+    print("RAISED-OK")
+s._backend.free_all()
+"""
+    env = dict(os.environ, DDSTORE_STRICT="1")
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300,
+                       cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr
+    assert "RAISED-OK" in r.stdout
