@@ -511,12 +511,15 @@ public:
             {(int64_t)scratch},
             at::TensorOptions().dtype(at::kByte).device(idx.device()));
         HIP_CHECK(hipMemsetAsync(tiles.data_ptr(), 0, scratch, stream()));
-        ddstore::gather_csr_fused(stream(), (const void* const*)v.d_peers,
-                                  v.d_prefix, v.d_elem_prefix, nparts_, v.d_goff,
-                                  idx.data_ptr<int64_t>(), nidx,
-                                  off.data_ptr<int64_t>(),
-                                  v.row_elems * v.itemsize, cap,
-                                  out.data_ptr(), v.d_oob, tiles.data_ptr());
+        ddstore::csr_scan(stream(), v.d_goff, v.prefix[nparts_],
+                          idx.data_ptr<int64_t>(), nidx,
+                          off.data_ptr<int64_t>(), v.d_oob, tiles.data_ptr());
+        ddstore::gather_csr(stream(), (const void* const*)v.d_peers, v.d_prefix,
+                            v.d_elem_prefix, nparts_, v.d_goff,
+                            idx.data_ptr<int64_t>(), nidx,
+                            off.data_ptr<int64_t>(),
+                            v.row_elems * v.itemsize, cap,
+                            out.data_ptr(), v.d_oob);
         v.n_gather += 1;
         v.rows_gathered += nidx;
         // bytes accounted via the device DDS_CTR_ELEMS counter (see query)
@@ -1166,6 +1169,48 @@ private:
     EpochFSM fsm_;
     std::map<std::string, HostVar> vars_;
 };
+
+// ===========================================================================
+// Cycle traversal of a permutation (host): the in-place chunked reshuffle
+// processes slots in cycle order so a bounded buffer suffices -- slot j_i is
+// written from slot j_{i+1}=perm[j_i]'s OLD row, and within a contiguous
+// cycle traversal every read slot is only overwritten at the same or a later
+// position. Returns (order[n], starts[ncycles+1]): `order` is the
+// concatenated cycle traversal, `starts` its cycle start positions
+// (starts[ncycles] = n). Single O(n) pointer-chasing walk.
+// ===========================================================================
+inline std::pair<at::Tensor, at::Tensor> cycle_order(const at::Tensor& perm) {
+    TORCH_CHECK(perm.scalar_type() == at::kLong && perm.is_contiguous() &&
+                    perm.device().is_cpu(),
+                "ddstore cycle_order: perm must be a contiguous int64 CPU tensor");
+    const int64_t n = perm.numel();
+    const int64_t* p = perm.data_ptr<int64_t>();
+    at::Tensor order_t = at::empty({n}, at::TensorOptions().dtype(at::kLong));
+    int64_t* order = order_t.data_ptr<int64_t>();
+    std::vector<bool> visited((size_t)n, false);
+    std::vector<int64_t> starts;
+    int64_t w = 0;
+    for (int64_t s0 = 0; s0 < n; ++s0) {
+        if (visited[(size_t)s0]) continue;
+        starts.push_back(w);
+        int64_t j = s0;
+        while (!visited[(size_t)j]) {
+            visited[(size_t)j] = true;
+            order[w++] = j;
+            j = p[j];
+            TORCH_CHECK(j >= 0 && j < n,
+                        "ddstore cycle_order: not a permutation (value out of range)");
+        }
+        TORCH_CHECK(j == s0,
+                    "ddstore cycle_order: not a permutation (duplicate value)");
+    }
+    starts.push_back(n);
+    at::Tensor starts_t = at::empty({(int64_t)starts.size()},
+                                    at::TensorOptions().dtype(at::kLong));
+    std::memcpy(starts_t.data_ptr<int64_t>(), starts.data(),
+                starts.size() * sizeof(int64_t));
+    return {order_t, starts_t};
+}
 
 } // namespace ddstore
 

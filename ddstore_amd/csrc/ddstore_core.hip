@@ -32,6 +32,8 @@ using ddstore::HostStore;
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "ddstore_amd native core (MI355X / gfx950)";
     m.attr("MAX_PARTS") = DDS_MAX_PARTS;
+    m.def("cycle_order", &ddstore::cycle_order,
+          "Concatenated-cycle traversal of a permutation (order, starts)");
 
     py::class_<DeviceStore>(m, "DeviceStore")
         .def(py::init<int, int, int>(), py::arg("device"), py::arg("rank"),

@@ -92,22 +92,19 @@ void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
               int64_t nidx, int64_t nsamples, int64_t* d_lens,
               unsigned long long* d_oob, unsigned long long* d_elems);
 
-// Fused CSR fetch: ONE kernel computes per-sample lengths, their exclusive
-// scan (decoupled lookback across workgroups), writes d_out_off[nidx+1] and
-// gathers every sample's payload -- replaces the lens+cumsum+gather pipeline
-// (~25 us of plan overhead at B=262144 in round 1). `d_tiles` is a zeroed
-// scratch block of csr_fused_scratch_bytes(nidx); d_ctrs as in gather_csr.
+// Fused CSR plan: ONE kernel computes per-sample lengths AND their
+// exclusive scan (decoupled lookback across workgroup tiles), writing
+// d_out_off[nidx+1] -- replaces the zeros+lens+cumsum pipeline (~25 us of
+// plan overhead at B=262144 in round 1; a full lens+scan+gather fusion was
+// measured 2x slower, see k_csr_scan comment). Also accumulates the
+// requested element total into d_ctrs[DDS_CTR_ELEMS] (true-bytes stats).
+// `d_tiles` is a zeroed scratch block of csr_fused_scratch_bytes(nidx).
 size_t csr_fused_scratch_bytes(int64_t nidx);
-void gather_csr_fused(hipStream_t stream,
-                      const void* const* d_peer_base,
-                      const int64_t* d_sample_prefix,
-                      const int64_t* d_elem_prefix, int nparts,
-                      const int64_t* d_goff,
-                      const int64_t* d_idx, int64_t nidx,
-                      int64_t* d_out_off,
-                      int64_t elem_bytes, int64_t cap_elems,
-                      void* d_out, unsigned long long* d_ctrs,
-                      void* d_tiles);
+void csr_scan(hipStream_t stream,
+              const int64_t* d_goff, int64_t nsamples,
+              const int64_t* d_idx, int64_t nidx,
+              int64_t* d_out_off, unsigned long long* d_ctrs,
+              void* d_tiles);
 
 
 // Scatter rows of a packed buffer into the local shard at arbitrary local row

@@ -312,7 +312,12 @@ k_gather_csr_wave(const void* const* peer_base,
         const int64_t len = goff[g + 1] - e0;
         const int64_t o0 = out_off[s];
         if (o0 < 0 || o0 + len > cap_elems) {  // undersized capacity buffer:
-            if (tid == 0) atomicAdd(ctrs + DDS_CTR_CAP, 1ull);  // never write OOB
+            if (tid == 0) {                    // never write OOB
+                atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
+                // keep the true-bytes counter honest (see k_csr_scan)
+                atomicAdd(ctrs + DDS_CTR_ELEMS,
+                          (unsigned long long)(-(long long)len));
+            }
             continue;
         }
         const int64_t nch = len * chunks_per_elem;
@@ -354,7 +359,15 @@ k_csr_lens(const int64_t* goff, const int64_t* idx, int64_t nidx, int64_t nsampl
 // Copy nd dwords src->dst by `nthreads` cooperating threads (tid strided),
 // with the STORE side aligned up to dwordx4 (store width is the side the
 // memory system cannot split -- same measured rule as the cast kernels):
-// head dwords to 16-B alignment, uint4 store body fed by dword loads, tail.
+// head dwords to 16-B store alignment, then a uint4-store body fed by
+// align(4) uint4 LOADS -- gfx950 global loads only require dword alignment
+// for dwordx4, so the load side also runs at 16 B/lane even though sample
+// payload starts are only 4-B-granular (the compiler would scalarize to
+// dword loads on targets without unaligned vector access).
+struct __attribute__((aligned(4))) DW4A4 {  // 16-B payload, 4-B alignment
+    uint32_t v[4];
+};
+
 __device__ __forceinline__ void copy_dwords_store16(
     uint32_t* __restrict__ dst, const uint32_t* __restrict__ src,
     int64_t nd, int tid, int nthreads) {
@@ -363,14 +376,15 @@ __device__ __forceinline__ void copy_dwords_store16(
     for (int64_t i = tid; i < h; i += nthreads) dst[i] = src[i];
     const int64_t nb = (nd - h) >> 2;
     uint4* d4 = reinterpret_cast<uint4*>(dst + h);
-    const uint32_t* s2 = src + h;
+    const DW4A4* s4 = reinterpret_cast<const DW4A4*>(src + h);
     for (int64_t i = tid; i < nb; i += nthreads) {
-        uint4 v;
-        v.x = s2[4 * i];
-        v.y = s2[4 * i + 1];
-        v.z = s2[4 * i + 2];
-        v.w = s2[4 * i + 3];
-        d4[i] = v;
+        const DW4A4 t = s4[i];  // one align(4) dwordx4 load on gfx950
+        uint4 o;
+        o.x = t.v[0];
+        o.y = t.v[1];
+        o.z = t.v[2];
+        o.w = t.v[3];
+        d4[i] = o;
     }
     for (int64_t i = h + (nb << 2) + tid; i < nd; i += nthreads) dst[i] = src[i];
 }
@@ -414,7 +428,11 @@ k_gather_csr_dw(const void* const* peer_base,
         const int64_t len = goff[g + 1] - e0;
         const int64_t o0 = out_off[s];
         if (o0 < 0 || o0 + len > cap_elems) {
-            if (tid == 0) atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
+            if (tid == 0) {
+                atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
+                atomicAdd(ctrs + DDS_CTR_ELEMS,
+                          (unsigned long long)(-(long long)len));
+            }
             continue;
         }
         copy_dwords_store16(out + o0 * dwords_per_elem,
@@ -424,65 +442,44 @@ k_gather_csr_dw(const void* const* peer_base,
 }
 
 // ---------------------------------------------------------------------------
-// Fused CSR fetch: per-sample lens + exclusive scan + gather in ONE kernel.
+// Fused CSR plan: per-sample lens + exclusive scan -> out_off in ONE kernel
+// with a decoupled lookback across workgroup tiles of kBlock samples.
 //
 // Round-1 measured the separate plan (zeros + lens kernel + torch cumsum) at
-// ~25 us/step (B=262144); a naive serial-lens fusion was tried and reverted
-// (latency-bound). This version keeps the scan fully parallel with a
-// decoupled-lookback across workgroup tiles of kBlock samples: each tile
-// block-scans its 256 lens, publishes its aggregate to `tile_state`,
-// looks back over earlier tiles for its exclusive base (values arrive long
-// before payload copies finish, so the chain pipelines under the gather),
-// then gathers its samples' payloads with GROUP lanes per sample.
+// ~25 us/step (B=262144); a serial-lens full fusion was tried and reverted
+// (latency-bound), and a lens+scan+GATHER single kernel was measured 2x
+// SLOWER than the pipeline (the block-wide phase barriers serialize the
+// payload copies against the scan; r2 A/B: 186 vs 98 us/step). Scan-only
+// fusion keeps the payload gather streaming in its own kernel while cutting
+// the plan to one small launch + an 8-KB memset.
 //
 // tile_state[t] encodes (value << 2) | flag, flag 1 = aggregate ready,
 // 2 = inclusive prefix ready; it must be zeroed before launch. The grid is
 // capped to the occupancy-resident block count so every spinning tile's
 // producer is guaranteed to be scheduled (forward progress).
 // ---------------------------------------------------------------------------
-template <typename T, int GROUP, bool DW16>
 __global__ void __launch_bounds__(kBlock)
-k_gather_csr_fused(const void* const* peer_base,
-                   const int64_t* sample_prefix, const int64_t* elem_prefix,
-                   int nparts, const int64_t* goff,
-                   const int64_t* idx, int64_t nidx,
-                   int64_t* __restrict__ out_off, int64_t chunks_per_elem,
-                   int64_t cap_elems, T* __restrict__ out,
-                   unsigned long long* ctrs,
-                   unsigned long long* __restrict__ tile_state) {
-    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
-    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
-    __shared__ const T* s_base[DDS_MAX_PARTS];
-    __shared__ int64_t s_g[kBlock];     // global sample id (-1 = oob)
-    __shared__ int64_t s_len[kBlock];   // sample length (elements)
-    __shared__ int64_t s_excl[kBlock];  // global exclusive output offset
+k_csr_scan(const int64_t* goff, int64_t nsamples,
+           const int64_t* idx, int64_t nidx,
+           int64_t* __restrict__ out_off,
+           unsigned long long* ctrs,
+           unsigned long long* __restrict__ tile_state) {
     __shared__ int64_t s_wsum[kBlock / 64];
     __shared__ int64_t s_tile_base;
-    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
-        s_sprefix[i] = sample_prefix[i];
-        s_eprefix[i] = elem_prefix[i];
-    }
-    for (int i = threadIdx.x; i < nparts; i += kBlock)
-        s_base[i] = reinterpret_cast<const T*>(peer_base[i]);
-    __syncthreads();
-
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
     for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const int64_t i = tile * kBlock + threadIdx.x;
-        int64_t g = -1, L = 0;
+        int64_t L = 0;
         if (i < nidx) {
-            g = idx[i];
-            if (g < 0 || g >= s_sprefix[nparts]) {
+            const int64_t g = idx[i];
+            if (g < 0 || g >= nsamples) {
                 atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
-                g = -1;
             } else {
                 L = goff[g + 1] - goff[g];
             }
         }
-        s_g[threadIdx.x] = g;
-        s_len[threadIdx.x] = L;
         // 64-wide inclusive wave scan of L, then cross-wave bases via LDS
         int64_t x = L;
         for (int off = 1; off < 64; off <<= 1) {
@@ -514,43 +511,14 @@ k_gather_csr_fused(const void* const* peer_base,
             atomicExch(&tile_state[tile],
                        ((unsigned long long)(excl + acc) << 2) | 2ull);
             s_tile_base = excl;
+            // true-bytes stats: requested elements; the gather kernels
+            // subtract the lens of capacity-skipped samples (rare path)
             atomicAdd(ctrs + DDS_CTR_ELEMS, (unsigned long long)acc);
             if (tile == 0) out_off[0] = 0;
         }
         __syncthreads();
-        const int64_t my_excl = s_tile_base + s_wsum[wave] + (x - L);
-        s_excl[threadIdx.x] = my_excl;
-        if (i < nidx) out_off[i + 1] = my_excl + L;
-        __syncthreads();
-        // gather phase: GROUP lanes cooperate per sample of this tile
-        const int nsamp = (int)(nidx - tile * kBlock < kBlock
-                                    ? nidx - tile * kBlock : (int64_t)kBlock);
-        const int gid = threadIdx.x / GROUP;
-        const int tid = threadIdx.x % GROUP;
-        constexpr int GPB = kBlock / GROUP;
-        for (int s = gid; s < nsamp; s += GPB) {
-            const int64_t g2 = s_g[s];
-            if (g2 < 0) continue;
-            const int64_t len = s_len[s];
-            const int64_t o0 = s_excl[s];
-            if (o0 + len > cap_elems) {
-                if (tid == 0) atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
-                continue;
-            }
-            const int p = owner_of(s_sprefix, nparts, g2);
-            const int64_t e0 = goff[g2];
-            const T* src = s_base[p] + (e0 - s_eprefix[p]) * chunks_per_elem;
-            T* dst = out + o0 * chunks_per_elem;
-            if constexpr (DW16) {
-                copy_dwords_store16(reinterpret_cast<uint32_t*>(dst),
-                                    reinterpret_cast<const uint32_t*>(src),
-                                    len * chunks_per_elem, tid, GROUP);
-            } else {
-                const int64_t nch = len * chunks_per_elem;
-                for (int64_t c = tid; c < nch; c += GROUP) dst[c] = src[c];
-            }
-        }
-        __syncthreads();  // LDS tiles reused next iteration
+        if (i < nidx) out_off[i + 1] = s_tile_base + s_wsum[wave] + x;
+        __syncthreads();  // s_wsum reused next tile iteration
     }
 }
 
@@ -837,91 +805,31 @@ size_t csr_fused_scratch_bytes(int64_t nidx) {
     return (size_t)(ntiles > 0 ? ntiles : 1) * sizeof(unsigned long long);
 }
 
-namespace {
-
-// Occupancy-resident grid cap for the fused kernel: the decoupled-lookback
-// spin requires every earlier tile's producer block to be scheduled, so the
-// grid must not exceed the number of simultaneously-resident blocks.
-template <typename T, int GROUP, bool DW16>
-int fused_resident_grid() {
-    static const int cached = [] {
+void csr_scan(hipStream_t stream,
+              const int64_t* d_goff, int64_t nsamples,
+              const int64_t* d_idx, int64_t nidx,
+              int64_t* d_out_off, unsigned long long* d_ctrs,
+              void* d_tiles) {
+    if (nidx == 0) return;
+    // the decoupled-lookback spin requires every earlier tile's producer
+    // block to be scheduled: cap the grid at the occupancy-resident count
+    static const int resident = [] {
         int dev = 0;
         (void)hipGetDevice(&dev);
         hipDeviceProp_t prop{};
         (void)hipGetDeviceProperties(&prop, dev);
         int occ = 0;
         (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &occ, reinterpret_cast<const void*>(&k_gather_csr_fused<T, GROUP, DW16>),
-            kBlock, 0);
+            &occ, reinterpret_cast<const void*>(&k_csr_scan), kBlock, 0);
         if (occ < 1) occ = 1;
         int sms = prop.multiProcessorCount > 0 ? prop.multiProcessorCount : 64;
         return occ * sms;
     }();
-    return cached;
-}
-
-template <typename T, int GROUP, bool DW16>
-void launch_csr_fused(hipStream_t stream, const void* const* pb,
-                      const int64_t* sp, const int64_t* ep, int np,
-                      const int64_t* goff, const int64_t* idx, int64_t nidx,
-                      int64_t* out_off, int64_t cpe, int64_t cap, void* out,
-                      unsigned long long* ctrs, unsigned long long* tiles) {
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
-    int grid = fused_resident_grid<T, GROUP, DW16>();
-    if ((int64_t)grid > ntiles) grid = (int)ntiles;
-    hipLaunchKernelGGL((k_gather_csr_fused<T, GROUP, DW16>), dim3(grid),
-                       dim3(kBlock), 0, stream, pb, sp, ep, np, goff, idx, nidx,
-                       out_off, cpe, cap, (T*)out, ctrs, tiles);
-}
-
-} // namespace
-
-void gather_csr_fused(hipStream_t stream,
-                      const void* const* d_peer_base,
-                      const int64_t* d_sample_prefix,
-                      const int64_t* d_elem_prefix, int nparts,
-                      const int64_t* d_goff,
-                      const int64_t* d_idx, int64_t nidx,
-                      int64_t* d_out_off,
-                      int64_t elem_bytes, int64_t cap_elems,
-                      void* d_out, unsigned long long* d_ctrs,
-                      void* d_tiles) {
-    if (nidx == 0) return;
-    static const int g_override = [] {
-        const char* e = getenv("DDSTORE_CSR_GROUP");
-        return e ? atoi(e) : 0;
-    }();
-    int group = nidx >= 16384 ? 16 : (nidx >= 4096 ? 64 : 256);
-    if (g_override) group = g_override;
-    if (group < 16) group = 16;  // fused tiles use >=16-lane sample groups
-    const uintptr_t oa = (uintptr_t)d_out;
-    auto* tiles = reinterpret_cast<unsigned long long*>(d_tiles);
-#define DDS_CSRF(T, div, DW)                                                         \
-    do {                                                                             \
-        if (group == 16)                                                             \
-            launch_csr_fused<T, 16, DW>(stream, d_peer_base, d_sample_prefix,        \
-                                        d_elem_prefix, nparts, d_goff, d_idx, nidx,  \
-                                        d_out_off, elem_bytes / div, cap_elems,      \
-                                        d_out, d_ctrs, tiles);                       \
-        else if (group == 64)                                                        \
-            launch_csr_fused<T, 64, DW>(stream, d_peer_base, d_sample_prefix,        \
-                                        d_elem_prefix, nparts, d_goff, d_idx, nidx,  \
-                                        d_out_off, elem_bytes / div, cap_elems,      \
-                                        d_out, d_ctrs, tiles);                       \
-        else                                                                         \
-            launch_csr_fused<T, 256, DW>(stream, d_peer_base, d_sample_prefix,       \
-                                         d_elem_prefix, nparts, d_goff, d_idx, nidx, \
-                                         d_out_off, elem_bytes / div, cap_elems,     \
-                                         d_out, d_ctrs, tiles);                      \
-    } while (0)
-    if (elem_bytes % 16 == 0 && oa % 16 == 0) {
-        DDS_CSRF(uint4, 16, false);
-    } else if (elem_bytes % 4 == 0 && oa % 4 == 0) {
-        DDS_CSRF(uint32_t, 4, true);
-    } else {
-        DDS_CSRF(uint8_t, 1, false);
-    }
-#undef DDS_CSRF
+    const int grid = (int)(ntiles < resident ? ntiles : (int64_t)resident);
+    hipLaunchKernelGGL(k_csr_scan, dim3(grid), dim3(kBlock), 0, stream,
+                       d_goff, nsamples, d_idx, nidx, d_out_off, d_ctrs,
+                       reinterpret_cast<unsigned long long*>(d_tiles));
 }
 
 void scatter_rows_local(hipStream_t stream,

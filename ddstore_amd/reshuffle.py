@@ -128,6 +128,102 @@ def _reshuffle_csr(store: DDStore, name: str, seed: int) -> None:
     store.add_csr(name, values, lens)
 
 
+def reshuffle_epoch_chunked(
+    store: DDStore, name: str, seed: int, max_chunk_bytes: int = 1 << 30
+) -> None:
+    """In-place chunked reshuffle: same slot-preserving result as
+    :func:`reshuffle_epoch` (slot j ends up holding the row previously at
+    slot perm[j]) but with O(chunk) transient device memory instead of ~2x
+    the shard -- required near HBM capacity (BASELINE config 4, ~250 GiB
+    shards on a 288 GB GPU).
+
+    Slots are processed in concatenated-CYCLE order (native
+    ``cycle_order``): the write to slot order[i] takes the OLD row of slot
+    order[i+1] = perm[order[i]], and within a contiguous cycle traversal a
+    slot is never read after the position that overwrites it, so chunks of
+    the traversal can be applied sequentially with one bounded buffer. The
+    one exception is each cycle's CLOSING write, which needs the cycle
+    head's original row -- a random permutation has only ~ln(n) cycles, so
+    all heads are saved up front. Reads are one-sided batched gathers (the
+    store's own xGMI path; random slots stripe all 7 links), a barrier
+    separates every chunk's reads from its writes.
+    """
+    if store._backend.epoch_active():
+        raise RuntimeError(
+            "ddstore reshuffle: cannot move data inside an open epoch "
+            "(concurrent gets would race); call epoch_end() first"
+        )
+    q = store.query(name)
+    if q["is_csr"]:
+        raise ValueError(
+            "ddstore reshuffle_epoch_chunked: fixed-stride variables only "
+            "(CSR shard sizes change; use reshuffle_epoch)"
+        )
+    from . import _C
+
+    dev = store.device
+    rank = store.rank
+    ntotal = int(q["nrows_total"])
+    disp = int(q["disp"])
+    itemsize = int(q["itemsize"])
+    dtype = store._meta(name)["dtype"]
+    p0, p1 = int(q["prefix"][rank]), int(q["prefix"][rank + 1])
+    row_bytes = max(disp * itemsize, 1)
+
+    g = torch.Generator(device=dev)
+    g.manual_seed(int(seed))
+    perm = torch.randperm(ntotal, generator=g, device=dev)
+    _check_perm_agreement(store, perm)
+    order, starts = _C.cycle_order(perm.cpu())
+    del perm
+
+    closing_pos = starts[1:] - 1  # traversal positions that close a cycle
+    closing_slot = order[closing_pos]
+    heads = order[starts[:-1]]  # cycle head slot, aligned with closing_slot
+    # save head rows for the closing writes THIS rank owns, before any write
+    close_mine = (closing_slot >= p0) & (closing_slot < p1)
+    my_heads = heads[close_mine]
+    head_rows = (
+        store.get_batch(name, my_heads, dtype=dtype)
+        if my_heads.numel()
+        else None
+    )
+    head_idx_of_slot = {
+        int(s): k for k, s in enumerate(closing_slot[close_mine].tolist())
+    }
+    if store.mode == "hip":
+        torch.cuda.synchronize(dev)
+    store.comm.barrier()
+
+    chunk = max(1, int(max_chunk_bytes) // row_bytes)
+    for a in range(0, ntotal, chunk):
+        b = min(a + chunk, ntotal)
+        W = order[a:b]
+        mine = (W >= p0) & (W < p1)
+        w_mine = W[mine]
+        if w_mine.numel():
+            pos = torch.arange(a, b, dtype=torch.int64)[mine]
+            is_close = torch.isin(pos, closing_pos)
+            src = order[torch.clamp(pos + 1, max=ntotal - 1)]
+            # closing writes take the saved head row; give them a harmless
+            # in-range source slot for the batched gather, then substitute
+            src = torch.where(is_close, w_mine, src)
+            buf = store.get_batch(name, src, dtype=dtype)
+            if is_close.any():
+                rows = is_close.nonzero(as_tuple=True)[0]
+                for r in rows.tolist():
+                    buf[r] = head_rows[head_idx_of_slot[int(w_mine[r])]]
+        if store.mode == "hip":
+            torch.cuda.synchronize(dev)
+        store.comm.barrier()  # ALL reads of this chunk precede ANY write
+        if w_mine.numel():
+            lidx = (w_mine - p0).to(dev) if store.mode == "hip" else w_mine - p0
+            store._backend.scatter_local(name, lidx.contiguous(), buf)
+        if store.mode == "hip":
+            torch.cuda.synchronize(dev)
+        store.comm.barrier()  # writes land before the next chunk's reads
+
+
 def expected_perm(ntotal: int, seed: int, device) -> torch.Tensor:
     """The permutation a reshuffle with ``seed`` applies (for tests)."""
     g = torch.Generator(device=device)
