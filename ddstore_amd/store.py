@@ -467,12 +467,24 @@ class DDStore:
         return _cm()
 
     # --------------------------------------------------------------- reshuffle
-    def reshuffle(self, name: str, seed: int) -> None:
-        """Epoch-level global data reshuffle over xGMI (all-to-all); see
-        :func:`ddstore_amd.reshuffle.reshuffle_epoch`."""
-        from .reshuffle import reshuffle_epoch
+    def reshuffle(self, name: str, seed: int,
+                  max_chunk_bytes: Optional[int] = None) -> None:
+        """Epoch-level global data reshuffle over xGMI.
 
-        reshuffle_epoch(self, name, seed)
+        Default: one all-to-all exchange of the whole variable
+        (:func:`ddstore_amd.reshuffle.reshuffle_epoch`; transiently ~2x the
+        shard). With ``max_chunk_bytes`` set: in-place cycle-order chunked
+        exchange with O(chunk) transient memory -- required near HBM
+        capacity (:func:`ddstore_amd.reshuffle.reshuffle_epoch_chunked`,
+        fixed-stride variables only)."""
+        if max_chunk_bytes is not None:
+            from .reshuffle import reshuffle_epoch_chunked
+
+            reshuffle_epoch_chunked(self, name, seed, max_chunk_bytes)
+        else:
+            from .reshuffle import reshuffle_epoch
+
+            reshuffle_epoch(self, name, seed)
 
     # ------------------------------------------------------------ checkpoint
     def dump(self, name: str, path: str) -> None:

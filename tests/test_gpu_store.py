@@ -525,3 +525,16 @@ s._backend.free_all()
                        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert r.returncode == 0, r.stderr
     assert "RAISED-OK" in r.stdout
+
+
+def test_chunked_reshuffle_gpu(store):
+    from ddstore_amd.reshuffle import expected_perm
+
+    n = 4096
+    arr = torch.arange(n, dtype=torch.float32).unsqueeze(1).repeat(1, 16)
+    store.add("chk", arr)
+    store.reshuffle("chk", seed=5, max_chunk_bytes=16 * 64 * 100)  # many chunks
+    perm = expected_perm(n, 5, store.device).cpu()
+    out = store.get_batch("chk", list(range(n)))
+    torch.cuda.synchronize()
+    assert torch.equal(out.cpu(), arr[perm])
