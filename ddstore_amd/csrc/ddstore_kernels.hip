@@ -368,6 +368,11 @@ struct __attribute__((aligned(4))) DW4A4 {  // 16-B payload, 4-B alignment
     uint32_t v[4];
 };
 
+// Copy-body variants for the CSR dword path (selected by DDSTORE_CSR_COPY,
+// A/B evidence in profiles/): 0 = dword loads feeding dwordx4 stores
+// (round-1 shape), 1 = align(4) dwordx4 loads + dwordx4 stores,
+// 2 = variant 1 with nontemporal stores (write-once output bypasses L2).
+template <int VAR>
 __device__ __forceinline__ void copy_dwords_store16(
     uint32_t* __restrict__ dst, const uint32_t* __restrict__ src,
     int64_t nd, int tid, int nthreads) {
@@ -376,15 +381,34 @@ __device__ __forceinline__ void copy_dwords_store16(
     for (int64_t i = tid; i < h; i += nthreads) dst[i] = src[i];
     const int64_t nb = (nd - h) >> 2;
     uint4* d4 = reinterpret_cast<uint4*>(dst + h);
-    const DW4A4* s4 = reinterpret_cast<const DW4A4*>(src + h);
-    for (int64_t i = tid; i < nb; i += nthreads) {
-        const DW4A4 t = s4[i];  // one align(4) dwordx4 load on gfx950
-        uint4 o;
-        o.x = t.v[0];
-        o.y = t.v[1];
-        o.z = t.v[2];
-        o.w = t.v[3];
-        d4[i] = o;
+    if constexpr (VAR == 0) {
+        const uint32_t* s2 = src + h;
+        for (int64_t i = tid; i < nb; i += nthreads) {
+            uint4 o;
+            o.x = s2[4 * i];
+            o.y = s2[4 * i + 1];
+            o.z = s2[4 * i + 2];
+            o.w = s2[4 * i + 3];
+            d4[i] = o;
+        }
+    } else {
+        const DW4A4* s4 = reinterpret_cast<const DW4A4*>(src + h);
+        for (int64_t i = tid; i < nb; i += nthreads) {
+            const DW4A4 t = s4[i];  // one align(4) dwordx4 load on gfx950
+            uint4 o;
+            o.x = t.v[0];
+            o.y = t.v[1];
+            o.z = t.v[2];
+            o.w = t.v[3];
+            if constexpr (VAR == 2) {
+                typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+                u32x4 ev = {o.x, o.y, o.z, o.w};
+                __builtin_nontemporal_store(
+                    ev, reinterpret_cast<u32x4*>(&d4[i]));
+            } else {
+                d4[i] = o;
+            }
+        }
     }
     for (int64_t i = h + (nb << 2) + tid; i < nd; i += nthreads) dst[i] = src[i];
 }
@@ -393,7 +417,7 @@ __device__ __forceinline__ void copy_dwords_store16(
 // addressing with 16-B-aligned stores. GROUP lanes cooperate per sample
 // (16 for small samples so many samples stay in flight per CU, 64 or the
 // whole block for larger payloads).
-template <int GROUP>
+template <int GROUP, int VAR>
 __global__ void __launch_bounds__(kBlock)
 k_gather_csr_dw(const void* const* peer_base,
                 const int64_t* sample_prefix, const int64_t* elem_prefix, int nparts,
@@ -435,9 +459,9 @@ k_gather_csr_dw(const void* const* peer_base,
             }
             continue;
         }
-        copy_dwords_store16(out + o0 * dwords_per_elem,
-                            s_base[p] + (e0 - s_eprefix[p]) * dwords_per_elem,
-                            len * dwords_per_elem, tid, GROUP);
+        copy_dwords_store16<VAR>(out + o0 * dwords_per_elem,
+                                 s_base[p] + (e0 - s_eprefix[p]) * dwords_per_elem,
+                                 len * dwords_per_elem, tid, GROUP);
     }
 }
 
@@ -495,26 +519,51 @@ k_csr_scan(const int64_t* goff, int64_t nsamples,
                 s_wsum[w] = acc;
                 acc += t;
             }
+            s_tile_base = acc;  // stash the tile aggregate for wave 0
             __threadfence();
             atomicExch(&tile_state[tile],
                        ((unsigned long long)acc << 2) | 1ull);
+        }
+        __syncthreads();
+        if (wave == 0) {
+            // WAVE-PARALLEL lookback: 64 predecessor tiles per round trip.
+            // A single-lane walk serializes on L2 atomic latency (~60 ns x
+            // ntiles: measured 63 us at 1024 tiles, r2); here lane l reads
+            // tile t-l, the wave sums aggregates back to the nearest
+            // INCLUSIVE entry, and only continues past a full window of
+            // bare aggregates -- the chain cost drops by ~64x.
+            const int64_t acc = s_tile_base;
             int64_t excl = 0;
-            for (int64_t t = tile - 1; t >= 0; --t) {
-                unsigned long long v;
-                do {
-                    v = atomicAdd(&tile_state[t], 0ull);  // atomic read
-                } while ((v & 3ull) == 0ull);
-                excl += (int64_t)(v >> 2);
-                if ((v & 3ull) == 2ull) break;
+            int64_t t = tile - 1;
+            while (t >= 0) {
+                const int64_t pos = t - lane;
+                unsigned long long v = 0;
+                if (pos >= 0) {
+                    do {
+                        v = atomicAdd(&tile_state[pos], 0ull);  // atomic read
+                    } while ((v & 3ull) == 0ull);
+                }
+                // nearest lane holding an inclusive prefix (if any)
+                const unsigned long long ball =
+                    __ballot((pos >= 0) && (v & 3ull) == 2ull);
+                const int incl_lane = ball ? (__ffsll((long long)ball) - 1) : 64;
+                int64_t c = (pos >= 0 && lane <= incl_lane) ? (int64_t)(v >> 2) : 0;
+                for (int off = 32; off > 0; off >>= 1) c += __shfl_down(c, off);
+                c = __shfl(c, 0);
+                excl += c;
+                if (incl_lane < 64) break;
+                t -= 64;
             }
-            __threadfence();
-            atomicExch(&tile_state[tile],
-                       ((unsigned long long)(excl + acc) << 2) | 2ull);
-            s_tile_base = excl;
-            // true-bytes stats: requested elements; the gather kernels
-            // subtract the lens of capacity-skipped samples (rare path)
-            atomicAdd(ctrs + DDS_CTR_ELEMS, (unsigned long long)acc);
-            if (tile == 0) out_off[0] = 0;
+            if (lane == 0) {
+                __threadfence();
+                atomicExch(&tile_state[tile],
+                           ((unsigned long long)(excl + acc) << 2) | 2ull);
+                s_tile_base = excl;
+                // true-bytes stats: requested elements; the gather kernels
+                // subtract the lens of capacity-skipped samples (rare path)
+                atomicAdd(ctrs + DDS_CTR_ELEMS, (unsigned long long)acc);
+                if (tile == 0) out_off[0] = 0;
+            }
         }
         __syncthreads();
         if (i < nidx) out_off[i + 1] = s_tile_base + s_wsum[wave] + x;
@@ -779,11 +828,22 @@ void gather_csr(hipStream_t stream,
         else if (group == 64) DDS_CSR_G(T, div, 64);                                 \
         else DDS_CSR_G(T, div, 256);                                                 \
     } while (0)
-#define DDS_CSR_DW(G)                                                                \
-    hipLaunchKernelGGL((k_gather_csr_dw<G>), dim3(grid), dim3(kBlock), 0, stream,    \
+    static const int copy_var = [] {
+        const char* e = getenv("DDSTORE_CSR_COPY");
+        int v = e ? atoi(e) : 1;
+        return (v >= 0 && v <= 2) ? v : 1;
+    }();
+#define DDS_CSR_DW_V(G, V)                                                           \
+    hipLaunchKernelGGL((k_gather_csr_dw<G, V>), dim3(grid), dim3(kBlock), 0, stream, \
                        d_peer_base, d_sample_prefix, d_elem_prefix, nparts, d_goff,  \
                        d_idx, nidx, d_out_off, elem_bytes / 4, cap_elems,            \
                        (uint32_t*)d_out, d_ctrs)
+#define DDS_CSR_DW(G)                                                                \
+    do {                                                                             \
+        if (copy_var == 0) DDS_CSR_DW_V(G, 0);                                       \
+        else if (copy_var == 2) DDS_CSR_DW_V(G, 2);                                  \
+        else DDS_CSR_DW_V(G, 1);                                                     \
+    } while (0)
     if (elem_bytes % 16 == 0 && oa % 16 == 0) {
         DDS_CSR_T(uint4, 16);
     } else if (elem_bytes % 4 == 0 && oa % 4 == 0) {
@@ -799,6 +859,7 @@ void gather_csr(hipStream_t stream,
 #undef DDS_CSR_G
 #undef DDS_CSR_T
 #undef DDS_CSR_DW
+#undef DDS_CSR_DW_V
 }
 size_t csr_fused_scratch_bytes(int64_t nidx) {
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;

@@ -37,3 +37,76 @@ def test_gather_bandwidth_floors():
     if csr["GBps"] < CSR_FLOOR:
         failures.append(f"csr: {csr['GBps']:.0f} < floor {CSR_FLOOR}")
     assert not failures, "\n".join(failures)
+
+
+def test_prefetch_overlap_floor():
+    """Config-5 regression floor: >=85% of the fetch time
+    must be hidden under the train step by the side-stream prefetcher at a
+    bench-shaped config. hidden = T_fetch - (T_combined - T_train)."""
+    import time
+
+    import torch
+
+    from ddstore_amd import DDStore, PrefetchLoader
+
+    dev = torch.device("cuda:0")
+    torch.cuda.set_device(dev)
+    rows, dim, batch, steps = 1 << 21, 128, 1 << 18, 24
+    store = DDStore(device=dev)
+    store.add("ov", torch.randn(rows, dim, device=dev))
+
+    model = torch.nn.Sequential(
+        torch.nn.Linear(dim, 1024), torch.nn.GELU(), torch.nn.Linear(1024, dim)
+    ).to(device=dev, dtype=torch.bfloat16)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-3)
+
+    def train_step(b):
+        opt.zero_grad(set_to_none=True)
+        torch.nn.functional.mse_loss(model(b), b).backward()
+        opt.step()
+
+    def timed(fn, n):
+        fn()  # warm
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            fn()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n
+
+    g = torch.Generator().manual_seed(7)
+    order = torch.cat([torch.randperm(rows, generator=g) for _ in range(
+        (steps + 2) * batch // rows + 1)])[: (steps + 2) * batch]
+
+    # T_train: train only, fixed resident batch
+    fixed = torch.randn(batch, dim, dtype=torch.bfloat16, device=dev)
+    t_train = timed(lambda: train_step(fixed), steps)
+
+    # T_fetch: fetch only (same kernel the loader issues)
+    buf = torch.empty(batch, dim, dtype=torch.bfloat16, device=dev)
+    idx_dev = order[:batch].to(dev)
+    store.get_batch("ov", idx_dev, out=buf)
+    t_fetch = timed(lambda: store.gather_into("ov", idx_dev, buf), steps)
+
+    # T_combined: prefetch loader feeding the train step
+    store.epoch_begin()
+    loader = PrefetchLoader(store, "ov", order, batch,
+                            out_dtype=torch.bfloat16, depth=3, drop_last=True)
+    it = iter(loader)
+    train_step(next(it))  # warm
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        train_step(next(it))
+    torch.cuda.synchronize()
+    t_comb = (time.perf_counter() - t0) / steps
+    store.epoch_end()
+
+    hidden = t_fetch - max(t_comb - t_train, 0.0)
+    frac = hidden / t_fetch
+    store.free()
+    assert frac >= 0.85, (
+        f"prefetch overlap regressed: only {frac:.1%} of the fetch is hidden "
+        f"(t_train={t_train*1e3:.2f}ms t_fetch={t_fetch*1e3:.2f}ms "
+        f"t_combined={t_comb*1e3:.2f}ms)"
+    )
