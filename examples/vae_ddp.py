@@ -69,6 +69,34 @@ def loss_function(recon_x, x, mu, logvar):
     return bce + kld
 
 
+def load_mnist(path, nsamples):
+    """Load real MNIST from a local mnist.npz or idx-ubyte files (VERDICT r1
+    #8: no network in this environment, so the data must be vendored by the
+    user; the reference downloads via torchvision, vae-ddp.py:216)."""
+    import gzip
+
+    if os.path.isfile(path) and path.endswith(".npz"):
+        z = np.load(path)
+        data = z["x_train"].astype(np.float32) / 255.0
+        labels = z["y_train"].astype(np.int64)
+    else:
+        def rd(name):
+            for n in (name, name + ".gz"):
+                f = os.path.join(path, n)
+                if os.path.exists(f):
+                    op = gzip.open if f.endswith(".gz") else open
+                    with op(f, "rb") as fh:
+                        return fh.read()
+            raise FileNotFoundError(f"{name}[.gz] not under {path}")
+        raw = rd("train-images-idx3-ubyte")
+        data = np.frombuffer(raw, dtype=np.uint8, offset=16).reshape(
+            -1, 28, 28).astype(np.float32) / 255.0
+        raw = rd("train-labels-idx1-ubyte")
+        labels = np.frombuffer(raw, dtype=np.uint8, offset=8).astype(np.int64)
+    n = min(nsamples, data.shape[0])
+    return np.ascontiguousarray(data[:n]), np.ascontiguousarray(labels[:n])
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--epochs", type=int, default=2)
@@ -79,6 +107,13 @@ def main():
                    help="dist backend override (default nccl on GPU; use gloo "
                         "to oversubscribe ranks on one GPU)")
     p.add_argument("--ddstore-width", type=int, default=None)
+    p.add_argument("--data-path", default=None,
+                   help="real MNIST: path to mnist.npz or to a directory "
+                        "with train-images-idx3-ubyte / train-labels-idx1-"
+                        "ubyte (optionally .gz); no network is used")
+    p.add_argument("--assert-improve", type=float, default=None,
+                   help="require final-epoch loss < first-epoch loss * X "
+                        "(convergence gate; exits nonzero otherwise)")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -96,13 +131,17 @@ def main():
     else:
         device = torch.device("cpu")
 
-    # synthetic MNIST: blobby per-class patterns so the VAE has structure to fit
-    rng = np.random.default_rng(0)  # same dataset on every rank
-    labels = rng.integers(0, 10, size=args.nsamples)
-    protos = rng.random((10, 784)).astype(np.float32)
-    data = protos[labels] * 0.8 + rng.random(
-        (args.nsamples, 784)).astype(np.float32) * 0.2
-    data = data.reshape(args.nsamples, 28, 28)
+    if args.data_path:
+        data, labels = load_mnist(args.data_path, args.nsamples)
+    else:
+        # synthetic MNIST: per-class prototype patterns + noise -- structured
+        # enough that the VAE loss demonstrably converges (gated below)
+        rng = np.random.default_rng(0)  # same dataset on every rank
+        labels = rng.integers(0, 10, size=args.nsamples)
+        protos = rng.random((10, 784)).astype(np.float32)
+        data = protos[labels] * 0.8 + rng.random(
+            (args.nsamples, 784)).astype(np.float32) * 0.2
+        data = data.reshape(args.nsamples, 28, 28)
 
     ds = DistDataset(data, labels, device=device if use_cuda else "cpu",
                      ddstore_width=args.ddstore_width)
@@ -114,6 +153,7 @@ def main():
     opt = torch.optim.Adam(model.parameters(), lr=1e-3)
 
     n = len(ds)
+    epoch_losses = []
     for epoch in range(args.epochs):
         # DistributedSampler-style global shuffle (reference vae-ddp.py:216)
         g = torch.Generator().manual_seed(epoch)
@@ -134,10 +174,19 @@ def main():
             total += loss.item()
             nb += 1
         ds.epoch_end()
+        per_sample = total / max(nb * args.batch_size, 1)
+        epoch_losses.append(per_sample)
         if rank == 0:
             print(f"epoch {epoch}: train loss/sample "
-                  f"{total / max(nb * args.batch_size, 1):.3f} ({nb} batches)")
+                  f"{per_sample:.3f} ({nb} batches)")
     ds.free()
+    if args.assert_improve is not None and len(epoch_losses) >= 2:
+        ok = epoch_losses[-1] < epoch_losses[0] * args.assert_improve
+        if rank == 0:
+            print(f"convergence gate: first {epoch_losses[0]:.3f} -> last "
+                  f"{epoch_losses[-1]:.3f} ({'PASS' if ok else 'FAIL'})")
+        if not ok:
+            sys.exit(3)
     if world > 1:
         dist.destroy_process_group()
 
