@@ -50,6 +50,8 @@ def parse_args():
                         "(forward+backward+optimizer, BASELINE config 5); "
                         "csr = variable-length (HydraGNN-style) sample fetch "
                         "(BASELINE config 3)")
+    p.add_argument("--csr-fixed-len", action="store_true",
+                   help="csr mode: constant-length samples (A/B probe)")
     p.add_argument("--store-dtype", choices=["f32", "bf16", "u8", "fp8"], default="f32",
                    help="shard storage dtype; non-f32 exercises the fused "
                         "expand-on-gather path (fp8 = OCP e4m3fn)")
@@ -180,9 +182,14 @@ def main():
     sdt = {"f32": torch.float32, "bf16": torch.bfloat16, "u8": torch.uint8,
            "fp8": torch.float8_e4m3fn}[args.store_dtype]
     if args.mode == "csr":
-        # variable-length samples: 16..2*dim-16 f32 elements (mean = dim)
+        # variable-length samples: 16..2*dim-16 f32 elements (mean = dim);
+        # --csr-fixed-len makes every sample exactly dim elements (an A/B
+        # probe separating variable-length overhead from raw copy speed)
         gcpu = torch.Generator().manual_seed(4321 + rank)
-        lens = torch.randint(16, 2 * dim - 16, (rows,), generator=gcpu)
+        if args.csr_fixed_len:
+            lens = torch.full((rows,), dim, dtype=torch.int64)
+        else:
+            lens = torch.randint(16, 2 * dim - 16, (rows,), generator=gcpu)
         nelems = int(lens.sum())
         shard = torch.randn(nelems, 1, dtype=torch.float32, device=tdev)
         store.add_csr("bench", shard, lens)

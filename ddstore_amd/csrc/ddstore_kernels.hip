@@ -520,9 +520,9 @@ k_csr_scan(const int64_t* goff, int64_t nsamples,
                 acc += t;
             }
             s_tile_base = acc;  // stash the tile aggregate for wave 0
-            __threadfence();
-            atomicExch(&tile_state[tile],
-                       ((unsigned long long)acc << 2) | 1ull);
+            __hip_atomic_store(&tile_state[tile],
+                               ((unsigned long long)acc << 2) | 1ull,
+                               __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
         }
         __syncthreads();
         if (wave == 0) {
@@ -539,9 +539,17 @@ k_csr_scan(const int64_t* goff, int64_t nsamples,
                 const int64_t pos = t - lane;
                 unsigned long long v = 0;
                 if (pos >= 0) {
-                    do {
-                        v = atomicAdd(&tile_state[pos], 0ull);  // atomic read
-                    } while ((v & 3ull) == 0ull);
+                    // agent-scope atomic LOAD: a plain cache-bypassing read
+                    // that coalesces across lanes -- an atomicAdd(...,0)
+                    // spin is 64 uncoalesced L2 RMWs per window and was
+                    // measured to keep the scan at ~57 us (r2)
+                    for (;;) {
+                        v = __hip_atomic_load(&tile_state[pos],
+                                              __ATOMIC_ACQUIRE,
+                                              __HIP_MEMORY_SCOPE_AGENT);
+                        if ((v & 3ull) != 0ull) break;
+                        __builtin_amdgcn_s_sleep(1);
+                    }
                 }
                 // nearest lane holding an inclusive prefix (if any)
                 const unsigned long long ball =
@@ -555,9 +563,9 @@ k_csr_scan(const int64_t* goff, int64_t nsamples,
                 t -= 64;
             }
             if (lane == 0) {
-                __threadfence();
-                atomicExch(&tile_state[tile],
-                           ((unsigned long long)(excl + acc) << 2) | 2ull);
+                __hip_atomic_store(&tile_state[tile],
+                                   ((unsigned long long)(excl + acc) << 2) | 2ull,
+                                   __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
                 s_tile_base = excl;
                 // true-bytes stats: requested elements; the gather kernels
                 // subtract the lens of capacity-skipped samples (rare path)
@@ -825,13 +833,14 @@ void gather_csr(hipStream_t stream,
     do {                                                                             \
         if (group == 8) DDS_CSR_G(T, div, 8);                                        \
         else if (group == 16) DDS_CSR_G(T, div, 16);                                  \
+        else if (group == 32) DDS_CSR_G(T, div, 32);                                 \
         else if (group == 64) DDS_CSR_G(T, div, 64);                                 \
         else DDS_CSR_G(T, div, 256);                                                 \
     } while (0)
     static const int copy_var = [] {
         const char* e = getenv("DDSTORE_CSR_COPY");
-        int v = e ? atoi(e) : 1;
-        return (v >= 0 && v <= 2) ? v : 1;
+        int v = e ? atoi(e) : 0;  // A/B r2: dword loads won (1.845 vs
+        return (v >= 0 && v <= 2) ? v : 0;  // 1.802/1.773 G samples/s)
     }();
 #define DDS_CSR_DW_V(G, V)                                                           \
     hipLaunchKernelGGL((k_gather_csr_dw<G, V>), dim3(grid), dim3(kBlock), 0, stream, \
@@ -851,6 +860,7 @@ void gather_csr(hipStream_t stream,
         // dwordx4 inside each sample's payload
         if (group == 8) DDS_CSR_DW(8);
         else if (group == 16) DDS_CSR_DW(16);
+        else if (group == 32) DDS_CSR_DW(32);
         else if (group == 64) DDS_CSR_DW(64);
         else DDS_CSR_DW(256);
     } else {
