@@ -595,20 +595,19 @@ public:
             return off;
         }
         const int64_t cap = out.numel() / std::max<int64_t>(v.row_elems, 1);
-        const size_t scratch = ddstore::csr_fused_scratch_bytes(nidx);
-        at::Tensor tiles = at::empty(
-            {(int64_t)scratch},
-            at::TensorOptions().dtype(at::kByte).device(idx.device()));
-        HIP_CHECK(hipMemsetAsync(tiles.data_ptr(), 0, scratch, stream()));
-        ddstore::csr_scan(stream(), v.d_goff, v.prefix[nparts_],
-                          idx.data_ptr<int64_t>(), nidx,
-                          off.data_ptr<int64_t>(), v.d_oob, tiles.data_ptr());
-        ddstore::gather_csr(stream(), (const void* const*)v.d_peers, v.d_prefix,
-                            v.d_elem_prefix, nparts_, v.d_goff,
-                            idx.data_ptr<int64_t>(), nidx,
-                            off.data_ptr<int64_t>(),
-                            v.row_elems * v.itemsize, cap,
-                            out.data_ptr(), v.d_oob);
+        const int64_t elem_bytes = v.row_elems * v.itemsize;
+        auto bopts = at::TensorOptions().dtype(at::kByte).device(idx.device());
+        at::Tensor scratch = at::empty(
+            {(int64_t)ddstore::csr_plan_scratch_bytes(nidx)}, bopts);
+        const int64_t desc_cap =
+            cap / ddstore::csr_item_elems(elem_bytes) + nidx;
+        at::Tensor desc = at::empty({desc_cap}, opts);
+        ddstore::gather_csr_balanced(
+            stream(), (const void* const*)v.d_peers, v.d_prefix,
+            v.d_elem_prefix, nparts_, v.d_goff, v.prefix[nparts_],
+            idx.data_ptr<int64_t>(), nidx, elem_bytes, cap,
+            off.data_ptr<int64_t>(), out.data_ptr(), v.d_oob,
+            scratch.data_ptr(), desc.data_ptr<int64_t>(), desc_cap);
         v.n_gather += 1;
         v.rows_gathered += nidx;
         // bytes accounted via the device DDS_CTR_ELEMS counter (see query)

@@ -92,19 +92,27 @@ void csr_lens(hipStream_t stream, const int64_t* d_goff, const int64_t* d_idx,
               int64_t nidx, int64_t nsamples, int64_t* d_lens,
               unsigned long long* d_oob, unsigned long long* d_elems);
 
-// Fused CSR plan: ONE kernel computes per-sample lengths AND their
-// exclusive scan (decoupled lookback across workgroup tiles), writing
-// d_out_off[nidx+1] -- replaces the zeros+lens+cumsum pipeline (~25 us of
-// plan overhead at B=262144 in round 1; a full lens+scan+gather fusion was
-// measured 2x slower, see k_csr_scan comment). Also accumulates the
-// requested element total into d_ctrs[DDS_CTR_ELEMS] (true-bytes stats).
-// `d_tiles` is a zeroed scratch block of csr_fused_scratch_bytes(nidx).
-size_t csr_fused_scratch_bytes(int64_t nidx);
-void csr_scan(hipStream_t stream,
-              const int64_t* d_goff, int64_t nsamples,
-              const int64_t* d_idx, int64_t nidx,
-              int64_t* d_out_off, unsigned long long* d_ctrs,
-              void* d_tiles);
+// Balanced one-call CSR fetch: three tiny plan kernels (per-tile lens+scan,
+// single-block tile-aggregate scan, finalize + work-item emission) followed
+// by an item-parallel gather. Samples are split into <=~256-B work items so
+// gather waves get near-uniform work (the per-sample kernel loses ~1.5x to
+// max-of-N-samples-per-wave imbalance; r2 fixed-len A/B). Writes
+// d_out_off[nidx+1]; accumulates true gathered elements into
+// d_ctrs[DDS_CTR_ELEMS]; over-capacity samples are skipped + counted.
+// d_scratch: csr_plan_scratch_bytes(nidx) bytes (no zeroing needed);
+// d_desc: int64 work-item descriptors, capacity
+// cap_elems / csr_item_elems(elem_bytes) + nidx entries.
+size_t csr_plan_scratch_bytes(int64_t nidx);
+int64_t csr_item_elems(int64_t elem_bytes);
+void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
+                         const int64_t* d_sample_prefix,
+                         const int64_t* d_elem_prefix, int nparts,
+                         const int64_t* d_goff, int64_t nsamples_total,
+                         const int64_t* d_idx, int64_t nidx,
+                         int64_t elem_bytes, int64_t cap_elems,
+                         int64_t* d_out_off, void* d_out,
+                         unsigned long long* d_ctrs,
+                         void* d_scratch, int64_t* d_desc, int64_t desc_cap);
 
 
 // Scatter rows of a packed buffer into the local shard at arbitrary local row

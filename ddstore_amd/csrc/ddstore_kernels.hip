@@ -466,116 +466,248 @@ k_gather_csr_dw(const void* const* peer_base,
 }
 
 // ---------------------------------------------------------------------------
-// Fused CSR plan: per-sample lens + exclusive scan -> out_off in ONE kernel
-// with a decoupled lookback across workgroup tiles of kBlock samples.
+// CSR plan: per-sample lens + exclusive scan -> out_off, PLUS a balanced
+// work-item list for the gather.
 //
-// Round-1 measured the separate plan (zeros + lens kernel + torch cumsum) at
-// ~25 us/step (B=262144); a serial-lens full fusion was tried and reverted
-// (latency-bound), and a lens+scan+GATHER single kernel was measured 2x
-// SLOWER than the pipeline (the block-wide phase barriers serialize the
-// payload copies against the scan; r2 A/B: 186 vs 98 us/step). Scan-only
-// fusion keeps the payload gather streaming in its own kernel while cutting
-// the plan to one small launch + an 8-KB memset.
+// History of this path (all measured at B=262144, ~512 B mean samples):
+//   r1: zeros + lens kernel + torch cumsum               ~25 us plan
+//   r2a: single fused lens+scan+gather kernel            2x SLOWER overall
+//        (block-wide phase barriers serialize the payload copies)
+//   r2b: decoupled-lookback scan kernel                  57-83 us (the
+//        cross-tile flag traffic dominates regardless of spin flavor)
+//   r2c (this): THREE tiny deterministic kernels, no cross-block waiting:
+//        k_csr_plan1  per-tile lens + block scan -> partial out_off,
+//                     per-tile (elems, items) aggregates
+//        k_csr_plan2  one block scans the tile aggregates -> bases, totals
+//        k_csr_plan3  finalize out_off (+tile base), capacity check, and
+//                     EMIT work items: samples are split into <=ITEM-element
+//                     pieces so every gather group gets near-uniform work
+//                     (static sample-per-group scheduling loses ~1.5x to
+//                     the max-of-4-samples-per-wave effect; fixed-len A/B
+//                     r2: 46 vs 75 us for the same mean bytes).
 //
-// tile_state[t] encodes (value << 2) | flag, flag 1 = aggregate ready,
-// 2 = inclusive prefix ready; it must be zeroed before launch. The grid is
-// capped to the occupancy-resident block count so every spinning tile's
-// producer is guaranteed to be scheduled (forward progress).
+// Item descriptor: (sample_index << 20) | piece_index (piece < 2^20 =>
+// samples up to 2^20 * ITEM elements; guarded in plan3).
+// Scratch layout (csr_plan_scratch_bytes): [aggs 2*ntiles][lens nidx]
+// [meta 2] int64 each; meta = {total_items, total_elems}.
 // ---------------------------------------------------------------------------
+constexpr int kItemBytes = 256;  // per work item (A/B via DDSTORE_CSR_ITEM)
+
+__device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
+                                                int64_t* s_wsum2, int lane,
+                                                int wave, int64_t* tile_tot) {
+    // inclusive block scan of (x, y) over kBlock threads (wave shfl + LDS)
+    for (int off = 1; off < 64; off <<= 1) {
+        int64_t ax = __shfl_up(x, off);
+        int64_t ay = __shfl_up(y, off);
+        if (lane >= off) {
+            x += ax;
+            y += ay;
+        }
+    }
+    if (lane == 63) {
+        s_wsum2[2 * wave] = x;
+        s_wsum2[2 * wave + 1] = y;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int64_t ax = 0, ay = 0;
+        for (int w = 0; w < kBlock / 64; ++w) {
+            int64_t tx = s_wsum2[2 * w], ty = s_wsum2[2 * w + 1];
+            s_wsum2[2 * w] = ax;
+            s_wsum2[2 * w + 1] = ay;
+            ax += tx;
+            ay += ty;
+        }
+        if (tile_tot) {
+            tile_tot[0] = ax;
+            tile_tot[1] = ay;
+        }
+    }
+    __syncthreads();
+    x += s_wsum2[2 * wave];
+    y += s_wsum2[2 * wave + 1];
+}
+
+template <bool COUNT_OOB>
+__device__ __forceinline__ int64_t load_len(const int64_t* goff,
+                                            int64_t nsamples,
+                                            const int64_t* idx, int64_t i,
+                                            int64_t nidx,
+                                            unsigned long long* ctrs) {
+    if (i >= nidx) return 0;
+    const int64_t g = idx[i];
+    if (g < 0 || g >= nsamples) {
+        if (COUNT_OOB) atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
+        return 0;
+    }
+    return goff[g + 1] - goff[g];
+}
+
 __global__ void __launch_bounds__(kBlock)
-k_csr_scan(const int64_t* goff, int64_t nsamples,
-           const int64_t* idx, int64_t nidx,
-           int64_t* __restrict__ out_off,
-           unsigned long long* ctrs,
-           unsigned long long* __restrict__ tile_state) {
-    __shared__ int64_t s_wsum[kBlock / 64];
-    __shared__ int64_t s_tile_base;
+k_csr_plan1(const int64_t* goff, int64_t nsamples,
+            const int64_t* idx, int64_t nidx, int64_t item_elems,
+            int64_t* __restrict__ lens_tmp,
+            int64_t* __restrict__ aggs, unsigned long long* ctrs) {
+    __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
+    __shared__ int64_t s_tot[2];
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
     for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const int64_t i = tile * kBlock + threadIdx.x;
-        int64_t L = 0;
-        if (i < nidx) {
-            const int64_t g = idx[i];
-            if (g < 0 || g >= nsamples) {
-                atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
-            } else {
-                L = goff[g + 1] - goff[g];
-            }
-        }
-        // 64-wide inclusive wave scan of L, then cross-wave bases via LDS
+        const int64_t L = load_len<true>(goff, nsamples, idx, i, nidx, ctrs);
         int64_t x = L;
-        for (int off = 1; off < 64; off <<= 1) {
-            int64_t y = __shfl_up(x, off);
-            if (lane >= off) x += y;
+        int64_t y = L > 0 ? (L + item_elems - 1) / item_elems : 0;
+        if (i < nidx) lens_tmp[i] = L;
+        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot);
+        // (out_off is written complete, with bases, by plan3)
+        if (threadIdx.x == 0) {
+            aggs[2 * tile] = s_tot[0];
+            aggs[2 * tile + 1] = s_tot[1];
         }
-        if (lane == 63) s_wsum[wave] = x;
+        __syncthreads();  // LDS reused next tile
+    }
+}
+
+__global__ void __launch_bounds__(kBlock)
+k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
+            int64_t* __restrict__ meta, unsigned long long* ctrs) {
+    // ONE block: running exclusive scan of the (elems, items) tile
+    // aggregates in chunks of kBlock; writes totals into meta
+    __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
+    __shared__ int64_t s_tot[2];
+    __shared__ int64_t s_carry[2];
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    if (threadIdx.x == 0) {
+        s_carry[0] = 0;
+        s_carry[1] = 0;
+    }
+    __syncthreads();
+    for (int64_t base = 0; base < ntiles; base += kBlock) {
+        const int64_t t = base + threadIdx.x;
+        int64_t x = t < ntiles ? aggs[2 * t] : 0;
+        int64_t y = t < ntiles ? aggs[2 * t + 1] : 0;
+        const int64_t mx = x, my = y;
+        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot);
+        if (t < ntiles) {
+            aggs[2 * t] = s_carry[0] + x - mx;      // exclusive base
+            aggs[2 * t + 1] = s_carry[1] + y - my;
+        }
         __syncthreads();
         if (threadIdx.x == 0) {
-            int64_t acc = 0;
-            for (int w = 0; w < kBlock / 64; ++w) {
-                int64_t t = s_wsum[w];
-                s_wsum[w] = acc;
-                acc += t;
-            }
-            s_tile_base = acc;  // stash the tile aggregate for wave 0
-            __hip_atomic_store(&tile_state[tile],
-                               ((unsigned long long)acc << 2) | 1ull,
-                               __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+            s_carry[0] += s_tot[0];
+            s_carry[1] += s_tot[1];
         }
         __syncthreads();
-        if (wave == 0) {
-            // WAVE-PARALLEL lookback: 64 predecessor tiles per round trip.
-            // A single-lane walk serializes on L2 atomic latency (~60 ns x
-            // ntiles: measured 63 us at 1024 tiles, r2); here lane l reads
-            // tile t-l, the wave sums aggregates back to the nearest
-            // INCLUSIVE entry, and only continues past a full window of
-            // bare aggregates -- the chain cost drops by ~64x.
-            const int64_t acc = s_tile_base;
-            int64_t excl = 0;
-            int64_t t = tile - 1;
-            while (t >= 0) {
-                const int64_t pos = t - lane;
-                unsigned long long v = 0;
-                if (pos >= 0) {
-                    // agent-scope atomic LOAD: a plain cache-bypassing read
-                    // that coalesces across lanes -- an atomicAdd(...,0)
-                    // spin is 64 uncoalesced L2 RMWs per window and was
-                    // measured to keep the scan at ~57 us (r2)
-                    for (;;) {
-                        v = __hip_atomic_load(&tile_state[pos],
-                                              __ATOMIC_ACQUIRE,
-                                              __HIP_MEMORY_SCOPE_AGENT);
-                        if ((v & 3ull) != 0ull) break;
-                        __builtin_amdgcn_s_sleep(1);
-                    }
+    }
+    if (threadIdx.x == 0) {
+        meta[0] = s_carry[1];  // total items
+        meta[1] = s_carry[0];  // total elems
+        // true-bytes stats: requested elements (plan3 subtracts lens of
+        // capacity-skipped samples)
+        atomicAdd(ctrs + DDS_CTR_ELEMS, (unsigned long long)s_carry[0]);
+    }
+}
+
+__global__ void __launch_bounds__(kBlock)
+k_csr_plan3(const int64_t* __restrict__ lens_tmp, int64_t nidx,
+            int64_t item_elems, int64_t cap_elems,
+            const int64_t* __restrict__ aggs,
+            int64_t* __restrict__ out_off,
+            int64_t* __restrict__ desc, int64_t desc_cap,
+            unsigned long long* ctrs) {
+    __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
+    for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+        const int64_t i = tile * kBlock + threadIdx.x;
+        const int64_t L = i < nidx ? lens_tmp[i] : 0;
+        int64_t x = L;
+        int64_t items = L > 0 ? (L + item_elems - 1) / item_elems : 0;
+        int64_t y = items;
+        block_scan_pair(x, y, s_wsum2, lane, wave, nullptr);
+        const int64_t e_base = aggs[2 * tile], i_base = aggs[2 * tile + 1];
+        if (i < nidx) {
+            const int64_t excl_e = e_base + x - L;   // global elem offset
+            const int64_t excl_i = i_base + y - items;
+            out_off[i + 1] = e_base + x;
+            if (i == 0) out_off[0] = 0;
+            if (L > 0) {
+                if (excl_e + L > cap_elems) {
+                    // over-capacity: no items emitted, never written OOB
+                    atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
+                    atomicAdd(ctrs + DDS_CTR_ELEMS,
+                              (unsigned long long)(-(long long)L));
+                } else if (items >= (1 << 20) || excl_i + items > desc_cap) {
+                    // pathological sample/overflow: counted, skipped
+                    atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
+                    atomicAdd(ctrs + DDS_CTR_ELEMS,
+                              (unsigned long long)(-(long long)L));
+                } else {
+                    for (int64_t k = 0; k < items; ++k)
+                        desc[excl_i + k] = (i << 20) | k;
                 }
-                // nearest lane holding an inclusive prefix (if any)
-                const unsigned long long ball =
-                    __ballot((pos >= 0) && (v & 3ull) == 2ull);
-                const int incl_lane = ball ? (__ffsll((long long)ball) - 1) : 64;
-                int64_t c = (pos >= 0 && lane <= incl_lane) ? (int64_t)(v >> 2) : 0;
-                for (int off = 32; off > 0; off >>= 1) c += __shfl_down(c, off);
-                c = __shfl(c, 0);
-                excl += c;
-                if (incl_lane < 64) break;
-                t -= 64;
-            }
-            if (lane == 0) {
-                __hip_atomic_store(&tile_state[tile],
-                                   ((unsigned long long)(excl + acc) << 2) | 2ull,
-                                   __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-                s_tile_base = excl;
-                // true-bytes stats: requested elements; the gather kernels
-                // subtract the lens of capacity-skipped samples (rare path)
-                atomicAdd(ctrs + DDS_CTR_ELEMS, (unsigned long long)acc);
-                if (tile == 0) out_off[0] = 0;
             }
         }
         __syncthreads();
-        if (i < nidx) out_off[i + 1] = s_tile_base + s_wsum[wave] + x;
-        __syncthreads();  // s_wsum reused next tile iteration
+    }
+}
+
+// Balanced CSR gather: GROUP lanes per work item (one <=ITEM-element piece
+// of one sample); near-uniform work per wave kills the max-of-N-samples
+// imbalance of the per-sample kernel. Item count is read from meta on
+// device (the host never syncs for it).
+template <int GROUP, bool DW>
+__global__ void __launch_bounds__(kBlock)
+k_gather_csr_items(const void* const* peer_base,
+                   const int64_t* sample_prefix, const int64_t* elem_prefix,
+                   int nparts, const int64_t* goff,
+                   const int64_t* idx,
+                   const int64_t* __restrict__ out_off,
+                   const int64_t* __restrict__ desc,
+                   const int64_t* __restrict__ meta,
+                   int64_t item_elems, int64_t units_per_elem,
+                   void* __restrict__ out_, unsigned long long* ctrs) {
+    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
+    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
+    __shared__ const char* s_base[DDS_MAX_PARTS];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
+        s_sprefix[i] = sample_prefix[i];
+        s_eprefix[i] = elem_prefix[i];
+    }
+    for (int i = threadIdx.x; i < nparts; i += kBlock)
+        s_base[i] = reinterpret_cast<const char*>(peer_base[i]);
+    __syncthreads();
+    const int64_t total = meta[0];
+    constexpr int GPB = kBlock / GROUP;
+    const int64_t first = (int64_t)blockIdx.x * GPB + threadIdx.x / GROUP;
+    const int64_t step = (int64_t)gridDim.x * GPB;
+    const int tid = threadIdx.x % GROUP;
+    for (int64_t j = first; j < total; j += step) {
+        const int64_t d = desc[j];
+        const int64_t i = d >> 20;
+        const int64_t k = d & ((1 << 20) - 1);
+        const int64_t g = idx[i];  // valid: plan3 only emits in-range samples
+        const int p = owner_of(s_sprefix, nparts, g);
+        const int64_t lo = k * item_elems;
+        const int64_t o0 = out_off[i];
+        const int64_t n = min(item_elems, (out_off[i + 1] - o0) - lo);
+        if constexpr (DW) {
+            const int64_t dpe = units_per_elem;  // dwords per element
+            copy_dwords_store16<0>(
+                reinterpret_cast<uint32_t*>(out_) + (o0 + lo) * dpe,
+                reinterpret_cast<const uint32_t*>(s_base[p]) +
+                    (goff[g] - s_eprefix[p] + lo) * dpe,
+                n * dpe, tid, GROUP);
+        } else {
+            const int64_t bpe = units_per_elem;  // bytes per element
+            char* dst = reinterpret_cast<char*>(out_) + (o0 + lo) * bpe;
+            const char* src = s_base[p] + (goff[g] - s_eprefix[p] + lo) * bpe;
+            for (int64_t c = tid; c < n * bpe; c += GROUP) dst[c] = src[c];
+        }
     }
 }
 
@@ -871,36 +1003,62 @@ void gather_csr(hipStream_t stream,
 #undef DDS_CSR_DW
 #undef DDS_CSR_DW_V
 }
-size_t csr_fused_scratch_bytes(int64_t nidx) {
+size_t csr_plan_scratch_bytes(int64_t nidx) {
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
-    return (size_t)(ntiles > 0 ? ntiles : 1) * sizeof(unsigned long long);
+    return (size_t)(2 * (ntiles > 0 ? ntiles : 1) + (nidx > 0 ? nidx : 1) + 2)
+           * sizeof(int64_t);
 }
 
-void csr_scan(hipStream_t stream,
-              const int64_t* d_goff, int64_t nsamples,
-              const int64_t* d_idx, int64_t nidx,
-              int64_t* d_out_off, unsigned long long* d_ctrs,
-              void* d_tiles) {
-    if (nidx == 0) return;
-    // the decoupled-lookback spin requires every earlier tile's producer
-    // block to be scheduled: cap the grid at the occupancy-resident count
-    static const int resident = [] {
-        int dev = 0;
-        (void)hipGetDevice(&dev);
-        hipDeviceProp_t prop{};
-        (void)hipGetDeviceProperties(&prop, dev);
-        int occ = 0;
-        (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &occ, reinterpret_cast<const void*>(&k_csr_scan), kBlock, 0);
-        if (occ < 1) occ = 1;
-        int sms = prop.multiProcessorCount > 0 ? prop.multiProcessorCount : 64;
-        return occ * sms;
+int64_t csr_item_elems(int64_t elem_bytes) {
+    static const int item_bytes = [] {
+        const char* e = getenv("DDSTORE_CSR_ITEM");
+        int v = e ? atoi(e) : kItemBytes;
+        return v >= 16 ? v : kItemBytes;
     }();
+    int64_t ie = item_bytes / (elem_bytes > 0 ? elem_bytes : 1);
+    return ie > 0 ? ie : 1;
+}
+
+void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
+                         const int64_t* d_sample_prefix,
+                         const int64_t* d_elem_prefix, int nparts,
+                         const int64_t* d_goff, int64_t nsamples_total,
+                         const int64_t* d_idx, int64_t nidx,
+                         int64_t elem_bytes, int64_t cap_elems,
+                         int64_t* d_out_off, void* d_out,
+                         unsigned long long* d_ctrs,
+                         void* d_scratch, int64_t* d_desc, int64_t desc_cap) {
+    if (nidx == 0) return;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
-    const int grid = (int)(ntiles < resident ? ntiles : (int64_t)resident);
-    hipLaunchKernelGGL(k_csr_scan, dim3(grid), dim3(kBlock), 0, stream,
-                       d_goff, nsamples, d_idx, nidx, d_out_off, d_ctrs,
-                       reinterpret_cast<unsigned long long*>(d_tiles));
+    int64_t* aggs = reinterpret_cast<int64_t*>(d_scratch);
+    int64_t* lens_tmp = aggs + 2 * ntiles;
+    int64_t* meta = lens_tmp + nidx;
+    const int64_t item_elems = csr_item_elems(elem_bytes);
+    const int g1 = (int)(ntiles < kMaxBlocks ? ntiles : kMaxBlocks);
+    hipLaunchKernelGGL(k_csr_plan1, dim3(g1), dim3(kBlock), 0, stream, d_goff,
+                       nsamples_total, d_idx, nidx, item_elems, lens_tmp,
+                       aggs, d_ctrs);
+    hipLaunchKernelGGL(k_csr_plan2, dim3(1), dim3(kBlock), 0, stream, aggs,
+                       ntiles, meta, d_ctrs);
+    hipLaunchKernelGGL(k_csr_plan3, dim3(g1), dim3(kBlock), 0, stream,
+                       lens_tmp, nidx, item_elems, cap_elems, aggs, d_out_off,
+                       d_desc, desc_cap, d_ctrs);
+    const uintptr_t oa = (uintptr_t)d_out;
+    const int64_t gi = (desc_cap + (kBlock / 16) - 1) / (kBlock / 16);
+    const int g2 = (int)(gi < kMaxBlocks ? gi : kMaxBlocks);
+    if (elem_bytes % 4 == 0 && oa % 4 == 0) {
+        hipLaunchKernelGGL((k_gather_csr_items<16, true>), dim3(g2),
+                           dim3(kBlock), 0, stream, d_peer_base,
+                           d_sample_prefix, d_elem_prefix, nparts, d_goff,
+                           d_idx, d_out_off, d_desc, meta, item_elems,
+                           elem_bytes / 4, d_out, d_ctrs);
+    } else {
+        hipLaunchKernelGGL((k_gather_csr_items<16, false>), dim3(g2),
+                           dim3(kBlock), 0, stream, d_peer_base,
+                           d_sample_prefix, d_elem_prefix, nparts, d_goff,
+                           d_idx, d_out_off, d_desc, meta, item_elems,
+                           elem_bytes, d_out, d_ctrs);
+    }
 }
 
 void scatter_rows_local(hipStream_t stream,
