@@ -529,25 +529,10 @@ __device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
     y += s_wsum2[2 * wave + 1];
 }
 
-template <bool COUNT_OOB>
-__device__ __forceinline__ int64_t load_len(const int64_t* goff,
-                                            int64_t nsamples,
-                                            const int64_t* idx, int64_t i,
-                                            int64_t nidx,
-                                            unsigned long long* ctrs) {
-    if (i >= nidx) return 0;
-    const int64_t g = idx[i];
-    if (g < 0 || g >= nsamples) {
-        if (COUNT_OOB) atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
-        return 0;
-    }
-    return goff[g + 1] - goff[g];
-}
-
 __global__ void __launch_bounds__(kBlock)
 k_csr_plan1(const int64_t* goff, int64_t nsamples,
             const int64_t* idx, int64_t nidx, int64_t item_elems,
-            int64_t* __restrict__ lens_tmp,
+            int64_t* __restrict__ lens_tmp, int64_t* __restrict__ e0_tmp,
             int64_t* __restrict__ aggs, unsigned long long* ctrs) {
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
     __shared__ int64_t s_tot[2];
@@ -556,10 +541,20 @@ k_csr_plan1(const int64_t* goff, int64_t nsamples,
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
     for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const int64_t i = tile * kBlock + threadIdx.x;
-        const int64_t L = load_len<true>(goff, nsamples, idx, i, nidx, ctrs);
+        int64_t L = 0, e0 = -1;
+        if (i < nidx) {
+            const int64_t g = idx[i];
+            if (g < 0 || g >= nsamples) {
+                atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
+            } else {
+                e0 = goff[g];
+                L = goff[g + 1] - e0;
+            }
+            lens_tmp[i] = L;
+            e0_tmp[i] = e0;  // global element start (src side), cached for plan3
+        }
         int64_t x = L;
         int64_t y = L > 0 ? (L + item_elems - 1) / item_elems : 0;
-        if (i < nidx) lens_tmp[i] = L;
         block_scan_pair(x, y, s_wsum2, lane, wave, s_tot);
         // (out_off is written complete, with bases, by plan3)
         if (threadIdx.x == 0) {
@@ -611,14 +606,29 @@ k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
     }
 }
 
+// plan3 emits SELF-CONTAINED item descriptors (2 int64 each) so the gather
+// does zero random metadata loads (per-item idx/goff/out_off/owner lookups
+// were measured to cost MORE than the imbalance they fixed: 110 vs 75 us):
+//   desc[2j]   = (peer << 56) | src_unit_offset   (units: dwords or bytes)
+//   desc[2j+1] = (n_units << 44) | dst_unit_offset
 __global__ void __launch_bounds__(kBlock)
-k_csr_plan3(const int64_t* __restrict__ lens_tmp, int64_t nidx,
-            int64_t item_elems, int64_t cap_elems,
+k_csr_plan3(const int64_t* __restrict__ lens_tmp,
+            const int64_t* __restrict__ e0_tmp, int64_t nidx,
+            const int64_t* sample_prefix, const int64_t* elem_prefix,
+            int nparts, const int64_t* idx,
+            int64_t item_elems, int64_t units_per_elem, int64_t cap_elems,
             const int64_t* __restrict__ aggs,
             int64_t* __restrict__ out_off,
             int64_t* __restrict__ desc, int64_t desc_cap,
             unsigned long long* ctrs) {
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
+    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
+    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
+    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
+        s_sprefix[i] = sample_prefix[i];
+        s_eprefix[i] = elem_prefix[i];
+    }
+    __syncthreads();
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
@@ -636,19 +646,28 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp, int64_t nidx,
             out_off[i + 1] = e_base + x;
             if (i == 0) out_off[0] = 0;
             if (L > 0) {
-                if (excl_e + L > cap_elems) {
-                    // over-capacity: no items emitted, never written OOB
-                    atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
-                    atomicAdd(ctrs + DDS_CTR_ELEMS,
-                              (unsigned long long)(-(long long)L));
-                } else if (items >= (1 << 20) || excl_i + items > desc_cap) {
-                    // pathological sample/overflow: counted, skipped
+                if (excl_e + L > cap_elems ||
+                    excl_i + items > desc_cap ||
+                    L * units_per_elem >= (int64_t(1) << 44)) {
+                    // over-capacity (or pathological): no items emitted,
+                    // never written out of bounds; counted
                     atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
                     atomicAdd(ctrs + DDS_CTR_ELEMS,
                               (unsigned long long)(-(long long)L));
                 } else {
-                    for (int64_t k = 0; k < items; ++k)
-                        desc[excl_i + k] = (i << 20) | k;
+                    const int64_t g = idx[i];
+                    const int64_t p = owner_of(s_sprefix, nparts, g);
+                    const int64_t src0 =
+                        (e0_tmp[i] - s_eprefix[p]) * units_per_elem;
+                    const int64_t dst0 = excl_e * units_per_elem;
+                    const int64_t iu = item_elems * units_per_elem;
+                    const int64_t nu = L * units_per_elem;
+                    for (int64_t k = 0; k < items; ++k) {
+                        const int64_t off = k * iu;
+                        const int64_t n = nu - off < iu ? nu - off : iu;
+                        desc[2 * (excl_i + k)] = (p << 56) | (src0 + off);
+                        desc[2 * (excl_i + k) + 1] = (n << 44) | (dst0 + off);
+                    }
                 }
             }
         }
@@ -656,28 +675,18 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp, int64_t nidx,
     }
 }
 
-// Balanced CSR gather: GROUP lanes per work item (one <=ITEM-element piece
-// of one sample); near-uniform work per wave kills the max-of-N-samples
-// imbalance of the per-sample kernel. Item count is read from meta on
-// device (the host never syncs for it).
+// Balanced CSR gather: GROUP lanes per work item (one <=ITEM-byte piece of
+// one sample, self-contained descriptor); near-uniform work per wave kills
+// the max-of-N-samples imbalance of the per-sample kernel, and the
+// coalesced 16-B descriptor read is the ONLY metadata access. Item count
+// is read from meta on device (the host never syncs for it).
 template <int GROUP, bool DW>
 __global__ void __launch_bounds__(kBlock)
-k_gather_csr_items(const void* const* peer_base,
-                   const int64_t* sample_prefix, const int64_t* elem_prefix,
-                   int nparts, const int64_t* goff,
-                   const int64_t* idx,
-                   const int64_t* __restrict__ out_off,
+k_gather_csr_items(const void* const* peer_base, int nparts,
                    const int64_t* __restrict__ desc,
                    const int64_t* __restrict__ meta,
-                   int64_t item_elems, int64_t units_per_elem,
                    void* __restrict__ out_, unsigned long long* ctrs) {
-    __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
-    __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
     __shared__ const char* s_base[DDS_MAX_PARTS];
-    for (int i = threadIdx.x; i <= nparts; i += kBlock) {
-        s_sprefix[i] = sample_prefix[i];
-        s_eprefix[i] = elem_prefix[i];
-    }
     for (int i = threadIdx.x; i < nparts; i += kBlock)
         s_base[i] = reinterpret_cast<const char*>(peer_base[i]);
     __syncthreads();
@@ -686,27 +695,23 @@ k_gather_csr_items(const void* const* peer_base,
     const int64_t first = (int64_t)blockIdx.x * GPB + threadIdx.x / GROUP;
     const int64_t step = (int64_t)gridDim.x * GPB;
     const int tid = threadIdx.x % GROUP;
+    (void)ctrs;
     for (int64_t j = first; j < total; j += step) {
-        const int64_t d = desc[j];
-        const int64_t i = d >> 20;
-        const int64_t k = d & ((1 << 20) - 1);
-        const int64_t g = idx[i];  // valid: plan3 only emits in-range samples
-        const int p = owner_of(s_sprefix, nparts, g);
-        const int64_t lo = k * item_elems;
-        const int64_t o0 = out_off[i];
-        const int64_t n = min(item_elems, (out_off[i + 1] - o0) - lo);
+        const int64_t d0 = desc[2 * j];
+        const int64_t d1 = desc[2 * j + 1];
+        const int p = (int)(d0 >> 56);
+        const int64_t src_u = d0 & ((int64_t(1) << 56) - 1);
+        const int64_t n_u = d1 >> 44;
+        const int64_t dst_u = d1 & ((int64_t(1) << 44) - 1);
         if constexpr (DW) {
-            const int64_t dpe = units_per_elem;  // dwords per element
             copy_dwords_store16<0>(
-                reinterpret_cast<uint32_t*>(out_) + (o0 + lo) * dpe,
-                reinterpret_cast<const uint32_t*>(s_base[p]) +
-                    (goff[g] - s_eprefix[p] + lo) * dpe,
-                n * dpe, tid, GROUP);
+                reinterpret_cast<uint32_t*>(out_) + dst_u,
+                reinterpret_cast<const uint32_t*>(s_base[p]) + src_u,
+                n_u, tid, GROUP);
         } else {
-            const int64_t bpe = units_per_elem;  // bytes per element
-            char* dst = reinterpret_cast<char*>(out_) + (o0 + lo) * bpe;
-            const char* src = s_base[p] + (goff[g] - s_eprefix[p] + lo) * bpe;
-            for (int64_t c = tid; c < n * bpe; c += GROUP) dst[c] = src[c];
+            char* dst = reinterpret_cast<char*>(out_) + dst_u;
+            const char* src = s_base[p] + src_u;
+            for (int64_t c = tid; c < n_u; c += GROUP) dst[c] = src[c];
         }
     }
 }
@@ -1005,8 +1010,8 @@ void gather_csr(hipStream_t stream,
 }
 size_t csr_plan_scratch_bytes(int64_t nidx) {
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
-    return (size_t)(2 * (ntiles > 0 ? ntiles : 1) + (nidx > 0 ? nidx : 1) + 2)
-           * sizeof(int64_t);
+    return (size_t)(2 * (ntiles > 0 ? ntiles : 1) + 2 * (nidx > 0 ? nidx : 1)
+                    + 2) * sizeof(int64_t);
 }
 
 int64_t csr_item_elems(int64_t elem_bytes) {
@@ -1032,32 +1037,32 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
     int64_t* aggs = reinterpret_cast<int64_t*>(d_scratch);
     int64_t* lens_tmp = aggs + 2 * ntiles;
-    int64_t* meta = lens_tmp + nidx;
+    int64_t* e0_tmp = lens_tmp + nidx;
+    int64_t* meta = e0_tmp + nidx;
     const int64_t item_elems = csr_item_elems(elem_bytes);
+    const uintptr_t oa = (uintptr_t)d_out;
+    const bool dw = elem_bytes % 4 == 0 && oa % 4 == 0;
+    const int64_t upe = dw ? elem_bytes / 4 : elem_bytes;  // units per elem
     const int g1 = (int)(ntiles < kMaxBlocks ? ntiles : kMaxBlocks);
     hipLaunchKernelGGL(k_csr_plan1, dim3(g1), dim3(kBlock), 0, stream, d_goff,
                        nsamples_total, d_idx, nidx, item_elems, lens_tmp,
-                       aggs, d_ctrs);
+                       e0_tmp, aggs, d_ctrs);
     hipLaunchKernelGGL(k_csr_plan2, dim3(1), dim3(kBlock), 0, stream, aggs,
                        ntiles, meta, d_ctrs);
     hipLaunchKernelGGL(k_csr_plan3, dim3(g1), dim3(kBlock), 0, stream,
-                       lens_tmp, nidx, item_elems, cap_elems, aggs, d_out_off,
-                       d_desc, desc_cap, d_ctrs);
-    const uintptr_t oa = (uintptr_t)d_out;
+                       lens_tmp, e0_tmp, nidx, d_sample_prefix, d_elem_prefix,
+                       nparts, d_idx, item_elems, upe, cap_elems, aggs,
+                       d_out_off, d_desc, desc_cap, d_ctrs);
     const int64_t gi = (desc_cap + (kBlock / 16) - 1) / (kBlock / 16);
     const int g2 = (int)(gi < kMaxBlocks ? gi : kMaxBlocks);
-    if (elem_bytes % 4 == 0 && oa % 4 == 0) {
+    if (dw) {
         hipLaunchKernelGGL((k_gather_csr_items<16, true>), dim3(g2),
-                           dim3(kBlock), 0, stream, d_peer_base,
-                           d_sample_prefix, d_elem_prefix, nparts, d_goff,
-                           d_idx, d_out_off, d_desc, meta, item_elems,
-                           elem_bytes / 4, d_out, d_ctrs);
+                           dim3(kBlock), 0, stream, d_peer_base, nparts,
+                           d_desc, meta, d_out, d_ctrs);
     } else {
         hipLaunchKernelGGL((k_gather_csr_items<16, false>), dim3(g2),
-                           dim3(kBlock), 0, stream, d_peer_base,
-                           d_sample_prefix, d_elem_prefix, nparts, d_goff,
-                           d_idx, d_out_off, d_desc, meta, item_elems,
-                           elem_bytes, d_out, d_ctrs);
+                           dim3(kBlock), 0, stream, d_peer_base, nparts,
+                           d_desc, meta, d_out, d_ctrs);
     }
 }
 
