@@ -314,7 +314,7 @@ k_gather_csr_wave(const void* const* peer_base,
         if (o0 < 0 || o0 + len > cap_elems) {  // undersized capacity buffer:
             if (tid == 0) {                    // never write OOB
                 atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
-                // keep the true-bytes counter honest (see k_csr_scan)
+                // keep the true-bytes counter honest (see k_csr_plan2)
                 atomicAdd(ctrs + DDS_CTR_ELEMS,
                           (unsigned long long)(-(long long)len));
             }
@@ -680,7 +680,7 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
 // the max-of-N-samples imbalance of the per-sample kernel, and the
 // coalesced 16-B descriptor read is the ONLY metadata access. Item count
 // is read from meta on device (the host never syncs for it).
-template <int GROUP, bool DW>
+template <int GROUP, bool DW, int VAR>
 __global__ void __launch_bounds__(kBlock)
 k_gather_csr_items(const void* const* peer_base, int nparts,
                    const int64_t* __restrict__ desc,
@@ -704,7 +704,7 @@ k_gather_csr_items(const void* const* peer_base, int nparts,
         const int64_t n_u = d1 >> 44;
         const int64_t dst_u = d1 & ((int64_t(1) << 44) - 1);
         if constexpr (DW) {
-            copy_dwords_store16<0>(
+            copy_dwords_store16<VAR>(
                 reinterpret_cast<uint32_t*>(out_) + dst_u,
                 reinterpret_cast<const uint32_t*>(s_base[p]) + src_u,
                 n_u, tid, GROUP);
@@ -1053,17 +1053,40 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
                        lens_tmp, e0_tmp, nidx, d_sample_prefix, d_elem_prefix,
                        nparts, d_idx, item_elems, upe, cap_elems, aggs,
                        d_out_off, d_desc, desc_cap, d_ctrs);
-    const int64_t gi = (desc_cap + (kBlock / 16) - 1) / (kBlock / 16);
+    static const int item_group = [] {
+        const char* e = getenv("DDSTORE_CSR_ITEM_GROUP");
+        int v = e ? atoi(e) : 16;
+        return (v == 8 || v == 16 || v == 32) ? v : 16;
+    }();
+    static const int item_var = [] {
+        const char* e = getenv("DDSTORE_CSR_COPY");
+        int v = e ? atoi(e) : 0;
+        return (v >= 0 && v <= 2) ? v : 0;
+    }();
+    const int64_t gi = (desc_cap + (kBlock / item_group) - 1)
+                       / (kBlock / item_group);
     const int g2 = (int)(gi < kMaxBlocks ? gi : kMaxBlocks);
+#define DDS_ITEMS_GV(G, V)                                                     \
+    hipLaunchKernelGGL((k_gather_csr_items<G, true, V>), dim3(g2),             \
+                       dim3(kBlock), 0, stream, d_peer_base, nparts, d_desc,   \
+                       meta, d_out, d_ctrs)
+#define DDS_ITEMS_G(G)                                                         \
+    do {                                                                       \
+        if (item_var == 1) DDS_ITEMS_GV(G, 1);                                 \
+        else if (item_var == 2) DDS_ITEMS_GV(G, 2);                            \
+        else DDS_ITEMS_GV(G, 0);                                               \
+    } while (0)
     if (dw) {
-        hipLaunchKernelGGL((k_gather_csr_items<16, true>), dim3(g2),
-                           dim3(kBlock), 0, stream, d_peer_base, nparts,
-                           d_desc, meta, d_out, d_ctrs);
+        if (item_group == 8) DDS_ITEMS_G(8);
+        else if (item_group == 32) DDS_ITEMS_G(32);
+        else DDS_ITEMS_G(16);
     } else {
-        hipLaunchKernelGGL((k_gather_csr_items<16, false>), dim3(g2),
+        hipLaunchKernelGGL((k_gather_csr_items<16, false, 0>), dim3(g2),
                            dim3(kBlock), 0, stream, d_peer_base, nparts,
                            d_desc, meta, d_out, d_ctrs);
     }
+#undef DDS_ITEMS_GV
+#undef DDS_ITEMS_G
 }
 
 void scatter_rows_local(hipStream_t stream,

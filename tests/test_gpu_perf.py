@@ -51,12 +51,15 @@ def test_prefetch_overlap_floor():
 
     dev = torch.device("cuda:0")
     torch.cuda.set_device(dev)
-    rows, dim, batch, steps = 1 << 21, 128, 1 << 18, 24
+    # sized so t_fetch is comparable to t_train: the hidden fraction is then
+    # resolvable against timing noise (a tiny fetch under a big step would
+    # pass trivially even with zero overlap)
+    rows, dim, batch, steps = 1 << 21, 128, 1 << 20, 30
     store = DDStore(device=dev)
     store.add("ov", torch.randn(rows, dim, device=dev))
 
     model = torch.nn.Sequential(
-        torch.nn.Linear(dim, 1024), torch.nn.GELU(), torch.nn.Linear(1024, dim)
+        torch.nn.Linear(dim, 512), torch.nn.GELU(), torch.nn.Linear(512, dim)
     ).to(device=dev, dtype=torch.bfloat16)
     opt = torch.optim.SGD(model.parameters(), lr=1e-3)
 
