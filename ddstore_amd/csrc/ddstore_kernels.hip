@@ -491,7 +491,12 @@ k_gather_csr_dw(const void* const* peer_base,
 // Scratch layout (csr_plan_scratch_bytes): [aggs 2*ntiles][lens nidx]
 // [meta 2] int64 each; meta = {total_items, total_elems}.
 // ---------------------------------------------------------------------------
-constexpr int kItemBytes = 256;  // per work item (A/B via DDSTORE_CSR_ITEM)
+// Work-item size: A/B on MI355X (B=262144, 16..240-elem f32 samples):
+// 256 B -> 2.39G, 512 B -> 2.65G, 1024 B -> 2.91G samples/s -- per-item
+// fixed cost beats the wave-balance gain, so items are sized to hold most
+// whole samples and splitting only kicks in for outliers (env override
+// DDSTORE_CSR_ITEM for other sample-size regimes).
+constexpr int kItemBytes = 1024;
 
 __device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
                                                 int64_t* s_wsum2, int lane,
@@ -529,13 +534,21 @@ __device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
     y += s_wsum2[2 * wave + 1];
 }
 
+__device__ void csr_aggs_scan_block(int64_t* __restrict__ aggs,
+                                    int64_t ntiles,
+                                    int64_t* __restrict__ meta,
+                                    unsigned long long* ctrs,
+                                    int64_t* s_wsum2);
+
 __global__ void __launch_bounds__(kBlock)
 k_csr_plan1(const int64_t* goff, int64_t nsamples,
             const int64_t* idx, int64_t nidx, int64_t item_elems,
             int64_t* __restrict__ lens_tmp, int64_t* __restrict__ e0_tmp,
-            int64_t* __restrict__ aggs, unsigned long long* ctrs) {
+            int64_t* __restrict__ aggs, int64_t* __restrict__ meta,
+            unsigned long long* ctrs) {
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
     __shared__ int64_t s_tot[2];
+    __shared__ int s_last;
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
@@ -547,8 +560,11 @@ k_csr_plan1(const int64_t* goff, int64_t nsamples,
             if (g < 0 || g >= nsamples) {
                 atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
             } else {
-                e0 = goff[g];
-                L = goff[g + 1] - e0;
+                // one align(8) 16-B load covers goff[g] and goff[g+1]
+                struct __attribute__((aligned(8))) LL2 { int64_t a, b; };
+                const LL2 v2 = *reinterpret_cast<const LL2*>(goff + g);
+                e0 = v2.a;
+                L = v2.b - v2.a;
             }
             lens_tmp[i] = L;
             e0_tmp[i] = e0;  // global element start (src side), cached for plan3
@@ -563,15 +579,29 @@ k_csr_plan1(const int64_t* goff, int64_t nsamples,
         }
         __syncthreads();  // LDS reused next tile
     }
+    // last finishing block converts the aggregates to exclusive bases
+    if (threadIdx.x == 0) {
+        __threadfence();  // publish this block's aggs device-wide
+        const unsigned long long done = atomicAdd(ctrs + DDS_CTR_PLAN, 1ull);
+        s_last = ((done + 1) % (unsigned long long)gridDim.x == 0) ? 1 : 0;
+    }
+    __syncthreads();
+    if (s_last) {
+        __threadfence();  // acquire every block's aggs
+        csr_aggs_scan_block(aggs, ntiles, meta, ctrs, s_wsum2);
+    }
 }
 
-__global__ void __launch_bounds__(kBlock)
-k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
-            int64_t* __restrict__ meta, unsigned long long* ctrs) {
-    // ONE block: running exclusive scan of the (elems, items) tile
-    // aggregates in chunks of kBlock; writes totals into meta
-    __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
-    __shared__ int64_t s_tot[2];
+// ONE block: running exclusive scan of the (elems, items) tile aggregates
+// in chunks of kBlock; writes totals into meta. Runs inline in plan1's
+// LAST finishing block (detected via the monotonic DDS_CTR_PLAN counter),
+// saving a launch gap per step.
+__device__ void csr_aggs_scan_block(int64_t* __restrict__ aggs,
+                                    int64_t ntiles,
+                                    int64_t* __restrict__ meta,
+                                    unsigned long long* ctrs,
+                                    int64_t* s_wsum2) {
+    __shared__ int64_t s_tot2[2];
     __shared__ int64_t s_carry[2];
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -585,15 +615,15 @@ k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
         int64_t x = t < ntiles ? aggs[2 * t] : 0;
         int64_t y = t < ntiles ? aggs[2 * t + 1] : 0;
         const int64_t mx = x, my = y;
-        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot);
+        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot2);
         if (t < ntiles) {
             aggs[2 * t] = s_carry[0] + x - mx;      // exclusive base
             aggs[2 * t + 1] = s_carry[1] + y - my;
         }
         __syncthreads();
         if (threadIdx.x == 0) {
-            s_carry[0] += s_tot[0];
-            s_carry[1] += s_tot[1];
+            s_carry[0] += s_tot2[0];
+            s_carry[1] += s_tot2[1];
         }
         __syncthreads();
     }
@@ -1046,9 +1076,7 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
     const int g1 = (int)(ntiles < kMaxBlocks ? ntiles : kMaxBlocks);
     hipLaunchKernelGGL(k_csr_plan1, dim3(g1), dim3(kBlock), 0, stream, d_goff,
                        nsamples_total, d_idx, nidx, item_elems, lens_tmp,
-                       e0_tmp, aggs, d_ctrs);
-    hipLaunchKernelGGL(k_csr_plan2, dim3(1), dim3(kBlock), 0, stream, aggs,
-                       ntiles, meta, d_ctrs);
+                       e0_tmp, aggs, meta, d_ctrs);
     hipLaunchKernelGGL(k_csr_plan3, dim3(g1), dim3(kBlock), 0, stream,
                        lens_tmp, e0_tmp, nidx, d_sample_prefix, d_elem_prefix,
                        nparts, d_idx, item_elems, upe, cap_elems, aggs,
