@@ -601,8 +601,7 @@ public:
             {(int64_t)ddstore::csr_plan_scratch_bytes(nidx)}, bopts);
         const int64_t desc_cap =
             cap / ddstore::csr_item_elems(elem_bytes) + nidx;
-        // 2 int64 per item x 4 size-band regions of desc_cap items each
-        at::Tensor desc = at::empty({2 * 4 * desc_cap}, opts);
+        at::Tensor desc = at::empty({2 * desc_cap}, opts);  // 2 i64 per item
         ddstore::gather_csr_balanced(
             stream(), (const void* const*)v.d_peers, v.d_prefix,
             v.d_elem_prefix, nparts_, v.d_goff, v.prefix[nparts_],

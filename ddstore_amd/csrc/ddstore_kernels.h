@@ -38,12 +38,7 @@ enum DDSType : int {
 #define DDS_CTR_OOB 0
 #define DDS_CTR_CAP 1
 #define DDS_CTR_ELEMS 2
-// monotonic plan-phase block counter (internal): every CSR plan launch adds
-// exactly gridDim blocks, so "(n+1) % grid == 0" fires once per launch --
-// the last finishing block runs the tile-aggregate scan inline (one launch
-// gap saved per step, no per-call zeroing needed)
-#define DDS_CTR_PLAN 3
-#define DDS_NCTR 4
+#define DDS_NCTR 3
 
 namespace ddstore {
 

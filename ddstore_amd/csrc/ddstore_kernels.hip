@@ -491,18 +491,17 @@ k_gather_csr_dw(const void* const* peer_base,
 // Scratch layout (csr_plan_scratch_bytes): [aggs 2*ntiles][lens nidx]
 // [meta 2] int64 each; meta = {total_items, total_elems}.
 // ---------------------------------------------------------------------------
-// Work-item size: A/B on MI355X (B=262144, 16..240-elem f32 samples):
-// 256 B -> 2.39G, 512 B -> 2.65G, 1024 B -> 2.91G samples/s -- per-item
-// fixed cost beats the wave-balance gain, so items are sized to hold most
-// whole samples and splitting only kicks in for outliers (env override
-// DDSTORE_CSR_ITEM for other sample-size regimes).
-constexpr int kItemBytes = 1024;
-
-// Work items are emitted into size BANDS (quartiles of the max item size):
-// the gather's waves then see near-uniform neighbors, recovering the
-// balance that per-sample scheduling loses to the max-of-N effect without
-// paying per-item metadata loads. Band regions are desc_cap items each.
-#define DDS_CSR_BANDS 4
+// Work-item size: A/B on MI355X (B=262144, 16..240-elem f32 samples), with
+// self-contained descriptors: 256 B -> 2.39G, 512 B -> 2.65G, 1024 B ->
+// 2.91G samples/s -- per-item fixed cost beats the wave-balance gain, so
+// items hold most whole samples and splitting only trims outliers.
+// Also tried and REVERTED (commit history has both sides): (a) fusing the
+// aggregate scan into plan1's last block -- the per-block __threadfence
+// costs an L2 writeback each, plan1 8 -> 41 us; (b) size-banded item
+// emission for wave uniformity -- the per-block same-address atomicAdd
+// allocation serializes at ~60 ns/op (plan3 6 -> 58 us) AND the banded
+// dst scatter loses write locality (gather 76 -> 85 us).
+constexpr int kItemBytes = 1024;  // env override: DDSTORE_CSR_ITEM
 
 __device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
                                                 int64_t* s_wsum2, int lane,
@@ -540,18 +539,11 @@ __device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
     y += s_wsum2[2 * wave + 1];
 }
 
-__device__ void csr_aggs_scan_block(int64_t* __restrict__ aggs,
-                                    int64_t ntiles,
-                                    int64_t* __restrict__ meta,
-                                    unsigned long long* ctrs,
-                                    int64_t* s_wsum2);
-
 __global__ void __launch_bounds__(kBlock)
 k_csr_plan1(const int64_t* goff, int64_t nsamples,
             const int64_t* idx, int64_t nidx, int64_t item_elems,
             int64_t* __restrict__ lens_tmp, int64_t* __restrict__ e0_tmp,
-            int64_t* __restrict__ aggs, int64_t* __restrict__ meta,
-            unsigned long long* ctrs) {
+            int64_t* __restrict__ aggs, unsigned long long* ctrs) {
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
     __shared__ int64_t s_tot[2];
     const int lane = threadIdx.x & 63;
@@ -565,11 +557,8 @@ k_csr_plan1(const int64_t* goff, int64_t nsamples,
             if (g < 0 || g >= nsamples) {
                 atomicAdd(ctrs + DDS_CTR_OOB, 1ull);
             } else {
-                // one align(8) 16-B load covers goff[g] and goff[g+1]
-                struct __attribute__((aligned(8))) LL2 { int64_t a, b; };
-                const LL2 v2 = *reinterpret_cast<const LL2*>(goff + g);
-                e0 = v2.a;
-                L = v2.b - v2.a;
+                e0 = goff[g];
+                L = goff[g + 1] - e0;
             }
             lens_tmp[i] = L;
             e0_tmp[i] = e0;  // global element start (src side), cached for plan3
@@ -584,24 +573,15 @@ k_csr_plan1(const int64_t* goff, int64_t nsamples,
         }
         __syncthreads();  // LDS reused next tile
     }
-    // NB: fusing the aggregate scan into the last finishing block here was
-    // tried and REVERTED: the per-block __threadfence() needed to publish
-    // aggs device-wide costs an L2 writeback per block and blew plan1 up
-    // from 8 to 41 us (r2 measured); the separate one-block kernel gets
-    // cross-block coherence from the launch boundary for free.
-    (void)meta;
 }
 
-// ONE block: running exclusive scan of the (elems, items) tile aggregates
-// in chunks of kBlock; writes totals into meta. Runs inline in plan1's
-// LAST finishing block (detected via the monotonic DDS_CTR_PLAN counter),
-// saving a launch gap per step.
-__device__ void csr_aggs_scan_block(int64_t* __restrict__ aggs,
-                                    int64_t ntiles,
-                                    int64_t* __restrict__ meta,
-                                    unsigned long long* ctrs,
-                                    int64_t* s_wsum2) {
-    __shared__ int64_t s_tot2[2];
+__global__ void __launch_bounds__(kBlock)
+k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
+            int64_t* __restrict__ meta, unsigned long long* ctrs) {
+    // ONE block: running exclusive scan of the (elems, items) tile
+    // aggregates in chunks of kBlock; writes totals into meta
+    __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
+    __shared__ int64_t s_tot[2];
     __shared__ int64_t s_carry[2];
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -615,15 +595,15 @@ __device__ void csr_aggs_scan_block(int64_t* __restrict__ aggs,
         int64_t x = t < ntiles ? aggs[2 * t] : 0;
         int64_t y = t < ntiles ? aggs[2 * t + 1] : 0;
         const int64_t mx = x, my = y;
-        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot2);
+        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot);
         if (t < ntiles) {
             aggs[2 * t] = s_carry[0] + x - mx;      // exclusive base
             aggs[2 * t + 1] = s_carry[1] + y - my;
         }
         __syncthreads();
         if (threadIdx.x == 0) {
-            s_carry[0] += s_tot2[0];
-            s_carry[1] += s_tot2[1];
+            s_carry[0] += s_tot[0];
+            s_carry[1] += s_tot[1];
         }
         __syncthreads();
     }
@@ -636,24 +616,11 @@ __device__ void csr_aggs_scan_block(int64_t* __restrict__ aggs,
     }
 }
 
-__global__ void __launch_bounds__(kBlock)
-k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
-            int64_t* __restrict__ meta, unsigned long long* ctrs) {
-    __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
-    if (threadIdx.x < DDS_CSR_BANDS) meta[2 + threadIdx.x] = 0;  // band counts
-    __syncthreads();
-    csr_aggs_scan_block(aggs, ntiles, meta, ctrs, s_wsum2);
-}
-
 // plan3 emits SELF-CONTAINED item descriptors (2 int64 each) so the gather
 // does zero random metadata loads (per-item idx/goff/out_off/owner lookups
 // were measured to cost MORE than the imbalance they fixed: 110 vs 75 us):
 //   desc[2j]   = (peer << 56) | src_unit_offset   (units: dwords or bytes)
 //   desc[2j+1] = (n_units << 44) | dst_unit_offset
-// Items land in one of DDS_CSR_BANDS size-quartile regions (desc_cap items
-// each, block-aggregated atomic placement) so gather waves see same-size
-// neighbors. Emitted samples' output ranges are disjoint within [0, cap],
-// so the total emitted item count always fits desc_cap.
 __global__ void __launch_bounds__(kBlock)
 k_csr_plan3(const int64_t* __restrict__ lens_tmp,
             const int64_t* __restrict__ e0_tmp, int64_t nidx,
@@ -663,12 +630,10 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
             const int64_t* __restrict__ aggs,
             int64_t* __restrict__ out_off,
             int64_t* __restrict__ desc, int64_t desc_cap,
-            int64_t* __restrict__ meta, unsigned long long* ctrs) {
+            unsigned long long* ctrs) {
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
     __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
     __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
-    __shared__ int64_t s_tot4[DDS_CSR_BANDS];
-    __shared__ int64_t s_bbase[DDS_CSR_BANDS];
     for (int i = threadIdx.x; i <= nparts; i += kBlock) {
         s_sprefix[i] = sample_prefix[i];
         s_eprefix[i] = elem_prefix[i];
@@ -676,23 +641,23 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
     __syncthreads();
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
-    const int64_t iu = item_elems * units_per_elem;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
     for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const int64_t i = tile * kBlock + threadIdx.x;
         const int64_t L = i < nidx ? lens_tmp[i] : 0;
         int64_t x = L;
-        const int64_t items = L > 0 ? (L + item_elems - 1) / item_elems : 0;
+        int64_t items = L > 0 ? (L + item_elems - 1) / item_elems : 0;
         int64_t y = items;
         block_scan_pair(x, y, s_wsum2, lane, wave, nullptr);
-        const int64_t e_base = aggs[2 * tile];
-        const int64_t excl_e = e_base + x - L;  // global elem offset
-        bool emit = false;
+        const int64_t e_base = aggs[2 * tile], i_base = aggs[2 * tile + 1];
         if (i < nidx) {
+            const int64_t excl_e = e_base + x - L;   // global elem offset
+            const int64_t excl_i = i_base + y - items;
             out_off[i + 1] = e_base + x;
             if (i == 0) out_off[0] = 0;
             if (L > 0) {
                 if (excl_e + L > cap_elems ||
+                    excl_i + items > desc_cap ||
                     L * units_per_elem >= (int64_t(1) << 44)) {
                     // over-capacity (or pathological): no items emitted,
                     // never written out of bounds; counted
@@ -700,49 +665,20 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
                     atomicAdd(ctrs + DDS_CTR_ELEMS,
                               (unsigned long long)(-(long long)L));
                 } else {
-                    emit = true;
+                    const int64_t g = idx[i];
+                    const int64_t p = owner_of(s_sprefix, nparts, g);
+                    const int64_t src0 =
+                        (e0_tmp[i] - s_eprefix[p]) * units_per_elem;
+                    const int64_t dst0 = excl_e * units_per_elem;
+                    const int64_t iu = item_elems * units_per_elem;
+                    const int64_t nu = L * units_per_elem;
+                    for (int64_t k = 0; k < items; ++k) {
+                        const int64_t off = k * iu;
+                        const int64_t n = nu - off < iu ? nu - off : iu;
+                        desc[2 * (excl_i + k)] = (p << 56) | (src0 + off);
+                        desc[2 * (excl_i + k) + 1] = (n << 44) | (dst0 + off);
+                    }
                 }
-            }
-        }
-        const int64_t nu = L * units_per_elem;
-        // band histogram of this thread's items (last piece may be ragged)
-        int64_t cnt[DDS_CSR_BANDS] = {0, 0, 0, 0};
-        if (emit) {
-            for (int64_t k = 0; k < items; ++k) {
-                const int64_t n = nu - k * iu < iu ? nu - k * iu : iu;
-                int b = (int)((n * DDS_CSR_BANDS) / (iu + 1));
-                cnt[b]++;
-            }
-        }
-        // block-exclusive offsets per band (two pair scans), then one
-        // global atomicAdd per band per block for the region bases
-        int64_t a0 = cnt[0], a1 = cnt[1], a2 = cnt[2], a3 = cnt[3];
-        block_scan_pair(a0, a1, s_wsum2, lane, wave, &s_tot4[0]);
-        block_scan_pair(a2, a3, s_wsum2, lane, wave, &s_tot4[2]);
-        if (threadIdx.x == 0) {
-            for (int b = 0; b < DDS_CSR_BANDS; ++b)
-                s_bbase[b] = s_tot4[b]
-                                 ? (int64_t)atomicAdd(
-                                       (unsigned long long*)&meta[2 + b],
-                                       (unsigned long long)s_tot4[b])
-                                 : 0;
-        }
-        __syncthreads();
-        if (emit) {
-            int64_t slot[DDS_CSR_BANDS] = {
-                s_bbase[0] + a0 - cnt[0], s_bbase[1] + a1 - cnt[1],
-                s_bbase[2] + a2 - cnt[2], s_bbase[3] + a3 - cnt[3]};
-            const int64_t g = idx[i];
-            const int64_t p = owner_of(s_sprefix, nparts, g);
-            const int64_t src0 = (e0_tmp[i] - s_eprefix[p]) * units_per_elem;
-            const int64_t dst0 = excl_e * units_per_elem;
-            for (int64_t k = 0; k < items; ++k) {
-                const int64_t off = k * iu;
-                const int64_t n = nu - off < iu ? nu - off : iu;
-                const int b = (int)((n * DDS_CSR_BANDS) / (iu + 1));
-                const int64_t j = b * desc_cap + slot[b]++;
-                desc[2 * j] = (p << 56) | (src0 + off);
-                desc[2 * j + 1] = (n << 44) | (dst0 + off);
             }
         }
         __syncthreads();
@@ -757,30 +693,22 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
 template <int GROUP, bool DW, int VAR>
 __global__ void __launch_bounds__(kBlock)
 k_gather_csr_items(const void* const* peer_base, int nparts,
-                   const int64_t* __restrict__ desc, int64_t desc_cap,
+                   const int64_t* __restrict__ desc,
                    const int64_t* __restrict__ meta,
                    void* __restrict__ out_, unsigned long long* ctrs) {
     __shared__ const char* s_base[DDS_MAX_PARTS];
     for (int i = threadIdx.x; i < nparts; i += kBlock)
         s_base[i] = reinterpret_cast<const char*>(peer_base[i]);
     __syncthreads();
-    // banded iteration: j walks the concatenation of the band regions so
-    // neighboring groups (one wave) handle same-quartile item sizes
-    int64_t pre[DDS_CSR_BANDS + 1];
-    pre[0] = 0;
-    for (int b = 0; b < DDS_CSR_BANDS; ++b) pre[b + 1] = pre[b] + meta[2 + b];
-    const int64_t total = pre[DDS_CSR_BANDS];
+    const int64_t total = meta[0];
     constexpr int GPB = kBlock / GROUP;
     const int64_t first = (int64_t)blockIdx.x * GPB + threadIdx.x / GROUP;
     const int64_t step = (int64_t)gridDim.x * GPB;
     const int tid = threadIdx.x % GROUP;
     (void)ctrs;
     for (int64_t j = first; j < total; j += step) {
-        int b = 0;
-        while (b + 1 < DDS_CSR_BANDS && j >= pre[b + 1]) ++b;
-        const int64_t jj = b * desc_cap + (j - pre[b]);
-        const int64_t d0 = desc[2 * jj];
-        const int64_t d1 = desc[2 * jj + 1];
+        const int64_t d0 = desc[2 * j];
+        const int64_t d1 = desc[2 * j + 1];
         const int p = (int)(d0 >> 56);
         const int64_t src_u = d0 & ((int64_t(1) << 56) - 1);
         const int64_t n_u = d1 >> 44;
@@ -1092,9 +1020,8 @@ void gather_csr(hipStream_t stream,
 }
 size_t csr_plan_scratch_bytes(int64_t nidx) {
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
-    // [aggs 2*ntiles][lens nidx][e0 nidx][meta: items, elems, 4 band counts]
     return (size_t)(2 * (ntiles > 0 ? ntiles : 1) + 2 * (nidx > 0 ? nidx : 1)
-                    + 2 + DDS_CSR_BANDS) * sizeof(int64_t);
+                    + 2) * sizeof(int64_t);
 }
 
 int64_t csr_item_elems(int64_t elem_bytes) {
@@ -1129,13 +1056,13 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
     const int g1 = (int)(ntiles < kMaxBlocks ? ntiles : kMaxBlocks);
     hipLaunchKernelGGL(k_csr_plan1, dim3(g1), dim3(kBlock), 0, stream, d_goff,
                        nsamples_total, d_idx, nidx, item_elems, lens_tmp,
-                       e0_tmp, aggs, meta, d_ctrs);
+                       e0_tmp, aggs, d_ctrs);
     hipLaunchKernelGGL(k_csr_plan2, dim3(1), dim3(kBlock), 0, stream, aggs,
                        ntiles, meta, d_ctrs);
     hipLaunchKernelGGL(k_csr_plan3, dim3(g1), dim3(kBlock), 0, stream,
                        lens_tmp, e0_tmp, nidx, d_sample_prefix, d_elem_prefix,
                        nparts, d_idx, item_elems, upe, cap_elems, aggs,
-                       d_out_off, d_desc, desc_cap, meta, d_ctrs);
+                       d_out_off, d_desc, desc_cap, d_ctrs);
     static const int item_group = [] {
         const char* e = getenv("DDSTORE_CSR_ITEM_GROUP");
         int v = e ? atoi(e) : 16;
@@ -1152,7 +1079,7 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
 #define DDS_ITEMS_GV(G, V)                                                     \
     hipLaunchKernelGGL((k_gather_csr_items<G, true, V>), dim3(g2),             \
                        dim3(kBlock), 0, stream, d_peer_base, nparts, d_desc,   \
-                       desc_cap, meta, d_out, d_ctrs)
+                       meta, d_out, d_ctrs)
 #define DDS_ITEMS_G(G)                                                         \
     do {                                                                       \
         if (item_var == 1) DDS_ITEMS_GV(G, 1);                                 \
@@ -1166,7 +1093,7 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
     } else {
         hipLaunchKernelGGL((k_gather_csr_items<16, false, 0>), dim3(g2),
                            dim3(kBlock), 0, stream, d_peer_base, nparts,
-                           d_desc, desc_cap, meta, d_out, d_ctrs);
+                           d_desc, meta, d_out, d_ctrs);
     }
 #undef DDS_ITEMS_GV
 #undef DDS_ITEMS_G
