@@ -113,3 +113,18 @@ def test_prefetch_overlap_floor():
         f"(t_train={t_train*1e3:.2f}ms t_fetch={t_fetch*1e3:.2f}ms "
         f"t_combined={t_comb*1e3:.2f}ms)"
     )
+
+
+def test_bench_mode_floors():
+    """End-to-end per-step floors for the two headline bench modes (r2
+    measured: fetch 5.97G, csr 2.9G samples/s at the default config;
+    floors leave ~12-15% for box variance)."""
+    for mode, floor in [("fetch", 5.2e9), ("csr", 2.5e9)]:
+        out = subprocess.run(
+            [sys.executable, "bench.py", "--steps", "300", "--warmup", "50",
+             "--mode", mode],
+            capture_output=True, text=True, timeout=300,
+        )
+        assert out.returncode == 0, out.stderr[-1500:]
+        val = json.loads(out.stdout.strip().splitlines()[-1])["value"]
+        assert val >= floor, f"{mode}: {val/1e9:.2f}G < floor {floor/1e9:.1f}G"
