@@ -1281,16 +1281,12 @@ private:
 // concatenated cycle traversal, `starts` its cycle start positions
 // (starts[ncycles] = n). Single O(n) pointer-chasing walk.
 // ===========================================================================
-inline std::pair<at::Tensor, at::Tensor> cycle_order(const at::Tensor& perm) {
-    TORCH_CHECK(perm.scalar_type() == at::kLong && perm.is_contiguous() &&
-                    perm.device().is_cpu(),
-                "ddstore cycle_order: perm must be a contiguous int64 CPU tensor");
-    const int64_t n = perm.numel();
-    const int64_t* p = perm.data_ptr<int64_t>();
-    at::Tensor order_t = at::empty({n}, at::TensorOptions().dtype(at::kLong));
-    int64_t* order = order_t.data_ptr<int64_t>();
+template <typename I>
+inline void cycle_walk(const I* p, int64_t n, int64_t* order,
+                       std::vector<int64_t>& starts) {
+    // the walk is a dependent random-access chain (unprefetchable); a
+    // 32-bit working copy halves its memory traffic for n < 2^31
     std::vector<bool> visited((size_t)n, false);
-    std::vector<int64_t> starts;
     int64_t w = 0;
     for (int64_t s0 = 0; s0 < n; ++s0) {
         if (visited[(size_t)s0]) continue;
@@ -1299,12 +1295,32 @@ inline std::pair<at::Tensor, at::Tensor> cycle_order(const at::Tensor& perm) {
         while (!visited[(size_t)j]) {
             visited[(size_t)j] = true;
             order[w++] = j;
-            j = p[j];
+            j = (int64_t)p[j];
             TORCH_CHECK(j >= 0 && j < n,
                         "ddstore cycle_order: not a permutation (value out of range)");
         }
         TORCH_CHECK(j == s0,
                     "ddstore cycle_order: not a permutation (duplicate value)");
+    }
+}
+
+inline std::pair<at::Tensor, at::Tensor> cycle_order(const at::Tensor& perm) {
+    TORCH_CHECK(perm.scalar_type() == at::kLong && perm.is_contiguous() &&
+                    perm.device().is_cpu(),
+                "ddstore cycle_order: perm must be a contiguous int64 CPU tensor");
+    const int64_t n = perm.numel();
+    const int64_t* p = perm.data_ptr<int64_t>();
+    at::Tensor order_t = at::empty({n}, at::TensorOptions().dtype(at::kLong));
+    int64_t* order = order_t.data_ptr<int64_t>();
+    std::vector<int64_t> starts;
+    if (n <= (int64_t)INT32_MAX) {
+        std::vector<int32_t> p32((size_t)n);
+        host_parallel_for(0, n, 1 << 20, [&](int64_t b, int64_t e) {
+            for (int64_t i = b; i < e; ++i) p32[(size_t)i] = (int32_t)p[i];
+        });
+        cycle_walk(p32.data(), n, order, starts);
+    } else {
+        cycle_walk(p, n, order, starts);
     }
     starts.push_back(n);
     at::Tensor starts_t = at::empty({(int64_t)starts.size()},

@@ -196,13 +196,21 @@ def reshuffle_epoch_chunked(
     store.comm.barrier()
 
     chunk = max(1, int(max_chunk_bytes) // row_bytes)
+    owns_all = p0 == 0 and p1 == ntotal  # world_size 1: skip the mask work
     for a in range(0, ntotal, chunk):
         b = min(a + chunk, ntotal)
         W = order[a:b]
-        mine = (W >= p0) & (W < p1)
-        w_mine = W[mine]
+        if owns_all:
+            w_mine = W
+        else:
+            mine = (W >= p0) & (W < p1)
+            w_mine = W[mine]
         if w_mine.numel():
-            pos = torch.arange(a, b, dtype=torch.int64)[mine]
+            pos = (
+                torch.arange(a, b, dtype=torch.int64)
+                if owns_all
+                else torch.arange(a, b, dtype=torch.int64)[mine]
+            )
             is_close = torch.isin(pos, closing_pos)
             src = order[torch.clamp(pos + 1, max=ntotal - 1)]
             # closing writes take the saved head row; give them a harmless
