@@ -229,11 +229,20 @@ def main():
         trainer = TrainStep(dim, args.hidden, device, world)
         loader = PrefetchLoader(store, "bench", order, batch,
                                 out_dtype=torch.bfloat16, depth=3, drop_last=True)
-        it = iter(loader)
+        state = {"it": iter(loader)}
+
+        def next_batch():
+            try:
+                return next(state["it"])
+            except StopIteration:
+                # index pool exhausted (capped at 64 Mi rows): cycle the
+                # same epoch order again -- long runs must not stop early
+                state["it"] = iter(loader)
+                return next(state["it"])
 
         def run_steps(n: int):
             for _ in range(n):
-                trainer(next(it))
+                trainer(next_batch())
 
     elif args.mode == "fetch":
         # flagship store metric: back-to-back batched gathers (owner lookup +

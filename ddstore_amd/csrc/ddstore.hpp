@@ -23,9 +23,11 @@
 #endif
 
 #include <atomic>
+#include <condition_variable>
 #include <cstdlib>
 #include <cstring>
 #include <exception>
+#include <functional>
 #include <map>
 #include <mutex>
 #include <string>
@@ -86,44 +88,105 @@ inline std::vector<int64_t> make_prefix(const std::vector<int64_t>& counts) {
 
 // Parallel-for for the HostStore hot loops. at::parallel_for from an
 // extension compiled without the wheel's OpenMP backend runs SEQUENTIALLY
-// (measured: 9 GB/s vs torch index_select's 57 on identical shm data, r2);
-// this spawns std::threads with dynamic range-stealing instead -- spawn
-// cost (~tens of us) is noise against the multi-ms gathers it serves, and
-// worker exceptions are rethrown in the caller (TORCH_CHECK-safe).
+// (measured: 9 GB/s vs torch index_select's 57 on identical shm data, r2).
+// A spawn-per-call std::thread version fixed the 8-core container but
+// collapsed on the 128-thread GPU-box host (127 thread spawns ~ the whole
+// gather); this PERSISTENT pool parks at::get_num_threads()-1 workers on a
+// condition variable and hands out ranges via an atomic cursor. Worker
+// exceptions are rethrown in the caller (TORCH_CHECK-safe); the singleton
+// is re-created after fork (pid check) and intentionally leaked.
+class HostPool {
+public:
+    static HostPool& inst() {
+        static std::mutex m;
+        static HostPool* p = nullptr;
+        static pid_t pid = 0;
+        std::lock_guard<std::mutex> l(m);
+        if (p == nullptr || pid != getpid()) {
+            p = new HostPool();  // leaked: no teardown-order races at exit
+            pid = getpid();
+        }
+        return *p;
+    }
+
+    void run(int64_t begin, int64_t end, int64_t step,
+             const std::function<void(int64_t, int64_t)>& f) {
+        std::unique_lock<std::mutex> lk(m_);
+        next_.store(begin, std::memory_order_relaxed);
+        end_ = end;
+        step_ = step;
+        fn_ = &f;
+        err_ = nullptr;
+        active_ = (int)workers_.size();
+        ++gen_;
+        cv_work_.notify_all();
+        lk.unlock();
+        work();  // the caller participates
+        lk.lock();
+        cv_done_.wait(lk, [&] { return active_ == 0; });
+        fn_ = nullptr;
+        if (err_) std::rethrow_exception(err_);
+    }
+
+private:
+    HostPool() {
+        const int n = std::max(1, (int)at::get_num_threads()) - 1;
+        workers_.reserve((size_t)n);
+        for (int i = 0; i < n; ++i)
+            workers_.emplace_back([this] { worker_loop(); });
+    }
+
+    void worker_loop() {
+        uint64_t seen = 0;
+        for (;;) {
+            std::unique_lock<std::mutex> lk(m_);
+            cv_work_.wait(lk, [&] { return gen_ != seen; });
+            seen = gen_;
+            lk.unlock();
+            work();
+            lk.lock();
+            if (--active_ == 0) cv_done_.notify_all();
+        }
+    }
+
+    void work() {
+        try {
+            for (;;) {
+                const int64_t b = next_.fetch_add(step_);
+                if (b >= end_) break;
+                (*fn_)(b, std::min(b + step_, end_));
+            }
+        } catch (...) {
+            std::lock_guard<std::mutex> l(em_);
+            if (!err_) err_ = std::current_exception();
+        }
+    }
+
+    std::mutex m_, em_;
+    std::condition_variable cv_work_, cv_done_;
+    std::vector<std::thread> workers_;
+    std::atomic<int64_t> next_{0};
+    int64_t end_ = 0, step_ = 1;
+    const std::function<void(int64_t, int64_t)>* fn_ = nullptr;
+    std::exception_ptr err_;
+    uint64_t gen_ = 0;
+    int active_ = 0;
+};
+
 template <typename F>
 inline void host_parallel_for(int64_t begin, int64_t end, int64_t grain,
                               const F& f) {
     const int64_t n = end - begin;
     if (n <= 0) return;
     if (grain < 1) grain = 1;
-    int nt = (int)std::min<int64_t>((int64_t)at::get_num_threads(),
-                                    (n + grain - 1) / grain);
-    if (nt <= 1) {
+    const int maxt = std::max(1, (int)at::get_num_threads());
+    if (n <= grain || maxt <= 1) {
         f(begin, end);
         return;
     }
-    std::atomic<int64_t> next(begin);
-    std::exception_ptr err = nullptr;
-    std::mutex em;
-    const int64_t step = std::max<int64_t>(grain, n / (4 * nt));
-    auto worker = [&]() {
-        try {
-            for (;;) {
-                int64_t b = next.fetch_add(step);
-                if (b >= end) break;
-                f(b, std::min(b + step, end));
-            }
-        } catch (...) {
-            std::lock_guard<std::mutex> l(em);
-            if (!err) err = std::current_exception();
-        }
-    };
-    std::vector<std::thread> ts;
-    ts.reserve(nt - 1);
-    for (int i = 1; i < nt; ++i) ts.emplace_back(worker);
-    worker();
-    for (auto& t : ts) t.join();
-    if (err) std::rethrow_exception(err);
+    const int64_t step = std::max<int64_t>(grain, n / (8 * maxt));
+    std::function<void(int64_t, int64_t)> fn(std::cref(f));
+    HostPool::inst().run(begin, end, step, fn);
 }
 
 // Row copy for the host gather path: the destination is written once and
@@ -984,6 +1047,9 @@ public:
             void* p = mmap(nullptr, nb, PROT_READ, MAP_SHARED, fd, 0);
             close(fd);
             TORCH_CHECK(p != MAP_FAILED, "ddstore: mmap failed for ", names[r]);
+#if defined(MADV_HUGEPAGE)
+            (void)madvise(p, nb, MADV_HUGEPAGE);
+#endif
             v.peers[r] = p;
             v.peer_bytes[r] = nb;
         }
@@ -1252,6 +1318,11 @@ private:
         v.base = mmap(nullptr, v.base_bytes, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
         close(fd);
         TORCH_CHECK(v.base != MAP_FAILED, "ddstore: mmap failed");
+#if defined(MADV_HUGEPAGE)
+        // shm defaults to 4-KB pages; where shmem THP is enabled ("advise")
+        // this collapses the random-gather TLB pressure
+        (void)madvise(v.base, v.base_bytes, MADV_HUGEPAGE);
+#endif
         v.creator = true;
     }
 
