@@ -57,7 +57,14 @@ class PrefetchLoader:
         depth: Optional[int] = None,
         drop_last: bool = False,
         affine: Optional[tuple] = None,
+        collect_events: bool = False,
     ):
+        # collect_events=True records (start, end) CUDA events around every
+        # side-stream fetch into self.fetch_events -- overlap instrumentation
+        # (the end-to-end wall-clock delta cannot resolve a ~100-us fetch
+        # under a multi-ms train step against timing noise)
+        self.collect_events = collect_events
+        self.fetch_events: list = []
         self.affine = affine
         if depth is None:
             depth = int(os.environ.get("DDSTORE_PREFETCH_DEPTH", "2"))
@@ -155,9 +162,16 @@ class PrefetchLoader:
             with torch.cuda.stream(side):
                 if slot["free_recorded"]:
                     side.wait_event(slot["free"])
+                if self.collect_events:
+                    ev_s = torch.cuda.Event(enable_timing=True)
+                    ev_s.record(side)
                 data, label = self._fetch(batches[j], slot)
                 slot["yield"] = data if label is None else (data, label)
                 slot["ready"].record(side)
+                if self.collect_events:
+                    ev_e = torch.cuda.Event(enable_timing=True)
+                    ev_e.record(side)
+                    self.fetch_events.append((ev_s, ev_e))
 
         for j in range(min(self.depth, nb)):
             launch(j)
