@@ -503,19 +503,6 @@ k_gather_csr_dw(const void* const* peer_base,
 // dst scatter loses write locality (gather 76 -> 85 us).
 constexpr int kItemBytes = 1024;  // env override: DDSTORE_CSR_ITEM
 
-// Items are emitted into DDS_CSR_NCLS size-class regions (quartiles of the
-// item size) with placement computed by SCANS -- fully deterministic, no
-// atomics, no fences (both measured to be the killers of the atomic-banded
-// attempt), and sample order is preserved WITHIN each class so dst writes
-// stay near-monotonic. Gather waves then see same-quartile neighbors,
-// recovering the wave balance that whole-sample items lose.
-#define DDS_CSR_NCLS 4
-
-__device__ __forceinline__ int item_class(int64_t n_e, int64_t item_elems) {
-    const int c = (int)((n_e * DDS_CSR_NCLS) / (item_elems + 1));
-    return c >= DDS_CSR_NCLS ? DDS_CSR_NCLS - 1 : c;
-}
-
 __device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
                                                 int64_t* s_wsum2, int lane,
                                                 int wave, int64_t* tile_tot) {
@@ -552,35 +539,13 @@ __device__ __forceinline__ void block_scan_pair(int64_t& x, int64_t& y,
     y += s_wsum2[2 * wave + 1];
 }
 
-// Scanned values per sample: [0] elems, [1..4] item count per size class.
-__device__ __forceinline__ void csr_plan_values(int64_t L, int64_t item_elems,
-                                                int64_t v[5]) {
-    v[0] = L;
-    v[1] = v[2] = v[3] = v[4] = 0;
-    for (int64_t off = 0; off < L; off += item_elems) {
-        const int64_t n = L - off < item_elems ? L - off : item_elems;
-        v[1 + item_class(n, item_elems)]++;
-    }
-}
-
-__device__ __forceinline__ void block_scan5(int64_t v[5], int64_t* s_wsum2,
-                                            int lane, int wave,
-                                            int64_t* tot6) {
-    int64_t z = 0;
-    block_scan_pair(v[0], v[1], s_wsum2, lane, wave,
-                    tot6 ? tot6 + 0 : nullptr);
-    block_scan_pair(v[2], v[3], s_wsum2, lane, wave,
-                    tot6 ? tot6 + 2 : nullptr);
-    block_scan_pair(v[4], z, s_wsum2, lane, wave, tot6 ? tot6 + 4 : nullptr);
-}
-
 __global__ void __launch_bounds__(kBlock)
 k_csr_plan1(const int64_t* goff, int64_t nsamples,
             const int64_t* idx, int64_t nidx, int64_t item_elems,
             int64_t* __restrict__ lens_tmp, int64_t* __restrict__ e0_tmp,
             int64_t* __restrict__ aggs, unsigned long long* ctrs) {
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
-    __shared__ int64_t s_tot[6];
+    __shared__ int64_t s_tot[2];
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
@@ -598,12 +563,14 @@ k_csr_plan1(const int64_t* goff, int64_t nsamples,
             lens_tmp[i] = L;
             e0_tmp[i] = e0;  // global element start (src side), cached for plan3
         }
-        int64_t v[5];
-        csr_plan_values(L, item_elems, v);
-        block_scan5(v, s_wsum2, lane, wave, s_tot);
+        int64_t x = L;
+        int64_t y = L > 0 ? (L + item_elems - 1) / item_elems : 0;
+        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot);
         // (out_off is written complete, with bases, by plan3)
-        if (threadIdx.x == 0)
-            for (int k = 0; k < 5; ++k) aggs[5 * tile + k] = s_tot[k];
+        if (threadIdx.x == 0) {
+            aggs[2 * tile] = s_tot[0];
+            aggs[2 * tile + 1] = s_tot[1];
+        }
         __syncthreads();  // LDS reused next tile
     }
 }
@@ -611,40 +578,38 @@ k_csr_plan1(const int64_t* goff, int64_t nsamples,
 __global__ void __launch_bounds__(kBlock)
 k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
             int64_t* __restrict__ meta, unsigned long long* ctrs) {
-    // ONE block: running exclusive scan of the per-tile (elems, class
-    // counts) aggregates in chunks of kBlock; writes totals + global class
-    // bases into meta = [total_items, total_elems, class_base[0..3]]
+    // ONE block: running exclusive scan of the (elems, items) tile
+    // aggregates in chunks of kBlock; writes totals into meta
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
-    __shared__ int64_t s_tot[6];
-    __shared__ int64_t s_carry[5];
+    __shared__ int64_t s_tot[2];
+    __shared__ int64_t s_carry[2];
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
-    if (threadIdx.x < 5) s_carry[threadIdx.x] = 0;
+    if (threadIdx.x == 0) {
+        s_carry[0] = 0;
+        s_carry[1] = 0;
+    }
     __syncthreads();
     for (int64_t base = 0; base < ntiles; base += kBlock) {
         const int64_t t = base + threadIdx.x;
-        int64_t v[5], m[5];
-        for (int k = 0; k < 5; ++k)
-            m[k] = v[k] = t < ntiles ? aggs[5 * t + k] : 0;
-        block_scan5(v, s_wsum2, lane, wave, s_tot);
-        if (t < ntiles)
-            for (int k = 0; k < 5; ++k)
-                aggs[5 * t + k] = s_carry[k] + v[k] - m[k];  // exclusive base
+        int64_t x = t < ntiles ? aggs[2 * t] : 0;
+        int64_t y = t < ntiles ? aggs[2 * t + 1] : 0;
+        const int64_t mx = x, my = y;
+        block_scan_pair(x, y, s_wsum2, lane, wave, s_tot);
+        if (t < ntiles) {
+            aggs[2 * t] = s_carry[0] + x - mx;      // exclusive base
+            aggs[2 * t + 1] = s_carry[1] + y - my;
+        }
         __syncthreads();
-        if (threadIdx.x == 0)
-            for (int k = 0; k < 5; ++k) s_carry[k] += s_tot[k];
+        if (threadIdx.x == 0) {
+            s_carry[0] += s_tot[0];
+            s_carry[1] += s_tot[1];
+        }
         __syncthreads();
     }
     if (threadIdx.x == 0) {
-        const int64_t items =
-            s_carry[1] + s_carry[2] + s_carry[3] + s_carry[4];
-        meta[0] = items;
+        meta[0] = s_carry[1];  // total items
         meta[1] = s_carry[0];  // total elems
-        int64_t b = 0;  // class regions are packed back to back
-        for (int c = 0; c < DDS_CSR_NCLS; ++c) {
-            meta[2 + c] = b;
-            b += s_carry[1 + c];
-        }
         // true-bytes stats: requested elements (plan3 subtracts lens of
         // capacity-skipped samples)
         atomicAdd(ctrs + DDS_CTR_ELEMS, (unsigned long long)s_carry[0]);
@@ -656,10 +621,6 @@ k_csr_plan2(int64_t* __restrict__ aggs, int64_t ntiles,
 // were measured to cost MORE than the imbalance they fixed: 110 vs 75 us):
 //   desc[2j]   = (peer << 56) | src_unit_offset   (units: dwords or bytes)
 //   desc[2j+1] = (n_units << 44) | dst_unit_offset
-// Placement is class-partitioned and scan-derived: item slot =
-// class_base (meta) + tile class base (aggs) + thread class exclusive.
-// Skipped samples (over-capacity) write NULL descriptors (n=0) into their
-// already-scanned slots so the packed regions stay gap-free.
 __global__ void __launch_bounds__(kBlock)
 k_csr_plan3(const int64_t* __restrict__ lens_tmp,
             const int64_t* __restrict__ e0_tmp, int64_t nidx,
@@ -667,19 +628,16 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
             int nparts, const int64_t* idx,
             int64_t item_elems, int64_t units_per_elem, int64_t cap_elems,
             const int64_t* __restrict__ aggs,
-            const int64_t* __restrict__ meta_ro,
             int64_t* __restrict__ out_off,
             int64_t* __restrict__ desc, int64_t desc_cap,
             unsigned long long* ctrs) {
     __shared__ int64_t s_wsum2[2 * (kBlock / 64)];
     __shared__ int64_t s_sprefix[DDS_MAX_PARTS + 1];
     __shared__ int64_t s_eprefix[DDS_MAX_PARTS + 1];
-    __shared__ int64_t s_cbase[DDS_CSR_NCLS];
     for (int i = threadIdx.x; i <= nparts; i += kBlock) {
         s_sprefix[i] = sample_prefix[i];
         s_eprefix[i] = elem_prefix[i];
     }
-    if (threadIdx.x < DDS_CSR_NCLS) s_cbase[threadIdx.x] = meta_ro[2 + threadIdx.x];
     __syncthreads();
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -687,50 +645,39 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
     for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const int64_t i = tile * kBlock + threadIdx.x;
         const int64_t L = i < nidx ? lens_tmp[i] : 0;
-        int64_t v[5];
-        csr_plan_values(L, item_elems, v);
-        const int64_t mine[5] = {v[0], v[1], v[2], v[3], v[4]};
-        block_scan5(v, s_wsum2, lane, wave, nullptr);
-        const int64_t e_base = aggs[5 * tile];
+        int64_t x = L;
+        int64_t items = L > 0 ? (L + item_elems - 1) / item_elems : 0;
+        int64_t y = items;
+        block_scan_pair(x, y, s_wsum2, lane, wave, nullptr);
+        const int64_t e_base = aggs[2 * tile], i_base = aggs[2 * tile + 1];
         if (i < nidx) {
-            const int64_t excl_e = e_base + v[0] - L;  // global elem offset
-            out_off[i + 1] = e_base + v[0];
+            const int64_t excl_e = e_base + x - L;   // global elem offset
+            const int64_t excl_i = i_base + y - items;
+            out_off[i + 1] = e_base + x;
             if (i == 0) out_off[0] = 0;
             if (L > 0) {
-                // next free slot per class for THIS sample's items
-                int64_t slot[DDS_CSR_NCLS];
-                int64_t max_slot = -1;
-                for (int c = 0; c < DDS_CSR_NCLS; ++c) {
-                    slot[c] = s_cbase[c] + aggs[5 * tile + 1 + c] +
-                              (v[1 + c] - mine[1 + c]);
-                    if (mine[1 + c] > 0)
-                        max_slot = max(max_slot, slot[c] + mine[1 + c] - 1);
-                }
-                const bool skip = excl_e + L > cap_elems ||
-                                  max_slot >= desc_cap ||
-                                  L * units_per_elem >= (int64_t(1) << 44);
-                if (skip) {
-                    // over-capacity (or pathological): never written out of
-                    // bounds; counted; slots become null descriptors
+                if (excl_e + L > cap_elems ||
+                    excl_i + items > desc_cap ||
+                    L * units_per_elem >= (int64_t(1) << 44)) {
+                    // over-capacity (or pathological): no items emitted,
+                    // never written out of bounds; counted
                     atomicAdd(ctrs + DDS_CTR_CAP, 1ull);
                     atomicAdd(ctrs + DDS_CTR_ELEMS,
                               (unsigned long long)(-(long long)L));
-                }
-                const int64_t g = idx[i];
-                const int64_t p = skip ? 0 : owner_of(s_sprefix, nparts, g);
-                const int64_t src0 =
-                    skip ? 0 : (e0_tmp[i] - s_eprefix[p]) * units_per_elem;
-                const int64_t dst0 = excl_e * units_per_elem;
-                const int64_t iu = item_elems * units_per_elem;
-                const int64_t nu = L * units_per_elem;
-                for (int64_t off = 0; off < nu; off += iu) {
-                    const int64_t n = nu - off < iu ? nu - off : iu;
-                    const int c = item_class(
-                        (n + units_per_elem - 1) / units_per_elem, item_elems);
-                    const int64_t j = slot[c]++;
-                    if (j >= desc_cap) continue;
-                    desc[2 * j] = skip ? 0 : ((p << 56) | (src0 + off));
-                    desc[2 * j + 1] = skip ? 0 : ((n << 44) | (dst0 + off));
+                } else {
+                    const int64_t g = idx[i];
+                    const int64_t p = owner_of(s_sprefix, nparts, g);
+                    const int64_t src0 =
+                        (e0_tmp[i] - s_eprefix[p]) * units_per_elem;
+                    const int64_t dst0 = excl_e * units_per_elem;
+                    const int64_t iu = item_elems * units_per_elem;
+                    const int64_t nu = L * units_per_elem;
+                    for (int64_t k = 0; k < items; ++k) {
+                        const int64_t off = k * iu;
+                        const int64_t n = nu - off < iu ? nu - off : iu;
+                        desc[2 * (excl_i + k)] = (p << 56) | (src0 + off);
+                        desc[2 * (excl_i + k) + 1] = (n << 44) | (dst0 + off);
+                    }
                 }
             }
         }
@@ -746,14 +693,14 @@ k_csr_plan3(const int64_t* __restrict__ lens_tmp,
 template <int GROUP, bool DW, int VAR>
 __global__ void __launch_bounds__(kBlock)
 k_gather_csr_items(const void* const* peer_base, int nparts,
-                   const int64_t* __restrict__ desc, int64_t desc_cap,
+                   const int64_t* __restrict__ desc,
                    const int64_t* __restrict__ meta,
                    void* __restrict__ out_, unsigned long long* ctrs) {
     __shared__ const char* s_base[DDS_MAX_PARTS];
     for (int i = threadIdx.x; i < nparts; i += kBlock)
         s_base[i] = reinterpret_cast<const char*>(peer_base[i]);
     __syncthreads();
-    const int64_t total = min(meta[0], desc_cap);
+    const int64_t total = meta[0];
     constexpr int GPB = kBlock / GROUP;
     const int64_t first = (int64_t)blockIdx.x * GPB + threadIdx.x / GROUP;
     const int64_t step = (int64_t)gridDim.x * GPB;
@@ -1073,9 +1020,8 @@ void gather_csr(hipStream_t stream,
 }
 size_t csr_plan_scratch_bytes(int64_t nidx) {
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
-    // [aggs 5*ntiles][lens nidx][e0 nidx][meta: items, elems, class bases]
-    return (size_t)(5 * (ntiles > 0 ? ntiles : 1) + 2 * (nidx > 0 ? nidx : 1)
-                    + 2 + DDS_CSR_NCLS) * sizeof(int64_t);
+    return (size_t)(2 * (ntiles > 0 ? ntiles : 1) + 2 * (nidx > 0 ? nidx : 1)
+                    + 2) * sizeof(int64_t);
 }
 
 int64_t csr_item_elems(int64_t elem_bytes) {
@@ -1100,7 +1046,7 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
     if (nidx == 0) return;
     const int64_t ntiles = (nidx + kBlock - 1) / kBlock;
     int64_t* aggs = reinterpret_cast<int64_t*>(d_scratch);
-    int64_t* lens_tmp = aggs + 5 * ntiles;
+    int64_t* lens_tmp = aggs + 2 * ntiles;
     int64_t* e0_tmp = lens_tmp + nidx;
     int64_t* meta = e0_tmp + nidx;
     const int64_t item_elems = csr_item_elems(elem_bytes);
@@ -1115,7 +1061,7 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
                        ntiles, meta, d_ctrs);
     hipLaunchKernelGGL(k_csr_plan3, dim3(g1), dim3(kBlock), 0, stream,
                        lens_tmp, e0_tmp, nidx, d_sample_prefix, d_elem_prefix,
-                       nparts, d_idx, item_elems, upe, cap_elems, aggs, meta,
+                       nparts, d_idx, item_elems, upe, cap_elems, aggs,
                        d_out_off, d_desc, desc_cap, d_ctrs);
     static const int item_group = [] {
         const char* e = getenv("DDSTORE_CSR_ITEM_GROUP");
@@ -1133,7 +1079,7 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
 #define DDS_ITEMS_GV(G, V)                                                     \
     hipLaunchKernelGGL((k_gather_csr_items<G, true, V>), dim3(g2),             \
                        dim3(kBlock), 0, stream, d_peer_base, nparts, d_desc,   \
-                       desc_cap, meta, d_out, d_ctrs)
+                       meta, d_out, d_ctrs)
 #define DDS_ITEMS_G(G)                                                         \
     do {                                                                       \
         if (item_var == 1) DDS_ITEMS_GV(G, 1);                                 \
@@ -1147,7 +1093,7 @@ void gather_csr_balanced(hipStream_t stream, const void* const* d_peer_base,
     } else {
         hipLaunchKernelGGL((k_gather_csr_items<16, false, 0>), dim3(g2),
                            dim3(kBlock), 0, stream, d_peer_base, nparts,
-                           d_desc, desc_cap, meta, d_out, d_ctrs);
+                           d_desc, meta, d_out, d_ctrs);
     }
 #undef DDS_ITEMS_GV
 #undef DDS_ITEMS_G
