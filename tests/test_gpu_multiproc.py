@@ -141,3 +141,23 @@ def _w_verify_transport_gpu(rank, world):
 
 def test_verify_transport_ipc():
     run_dist(_w_verify_transport_gpu, 2)
+
+
+def _w_chunked_reshuffle_gpu(rank, world):
+    from ddstore_amd import DDStore
+    from ddstore_amd.reshuffle import expected_perm, reshuffle_epoch_chunked
+
+    s = DDStore(device=_dev(rank))
+    base = torch.arange(rank * NUM, (rank + 1) * NUM, dtype=torch.float32)
+    s.add("x", base.unsqueeze(1).repeat(1, DIM))
+    # many small chunks: cross-rank one-sided reads + per-chunk barriers
+    reshuffle_epoch_chunked(s, "x", seed=29, max_chunk_bytes=NUM * DIM)
+    perm = expected_perm(NUM * world, 29, s.device)
+    out = s.get_batch("x", list(range(NUM * world)))
+    torch.cuda.synchronize()
+    assert torch.equal(out.cpu()[:, 0], perm.cpu().to(torch.float32))
+    s.free()
+
+
+def test_chunked_reshuffle_gpu_ws2():
+    run_dist(_w_chunked_reshuffle_gpu, 2)
