@@ -111,6 +111,9 @@ public:
 
     void run(int64_t begin, int64_t end, int64_t step,
              const std::function<void(int64_t, int64_t)>& f) {
+        // one task at a time: a second caller (e.g. another Python thread
+        // gathering concurrently) must not overwrite the live task state
+        std::lock_guard<std::mutex> excl(run_mx_);
         std::unique_lock<std::mutex> lk(m_);
         next_.store(begin, std::memory_order_relaxed);
         end_ = end;
@@ -162,7 +165,7 @@ private:
         }
     }
 
-    std::mutex m_, em_;
+    std::mutex run_mx_, m_, em_;
     std::condition_variable cv_work_, cv_done_;
     std::vector<std::thread> workers_;
     std::atomic<int64_t> next_{0};

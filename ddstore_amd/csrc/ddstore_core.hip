@@ -33,31 +33,32 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "ddstore_amd native core (MI355X / gfx950)";
     m.attr("MAX_PARTS") = DDS_MAX_PARTS;
     m.def("cycle_order", &ddstore::cycle_order,
+          py::call_guard<py::gil_scoped_release>(),
           "Concatenated-cycle traversal of a permutation (order, starts)");
 
     py::class_<DeviceStore>(m, "DeviceStore")
         .def(py::init<int, int, int>(), py::arg("device"), py::arg("rank"),
              py::arg("nparts"))
-        .def("add", &DeviceStore::add)
-        .def("init", &DeviceStore::init)
-        .def("add_csr", &DeviceStore::add_csr)
-        .def("init_csr", &DeviceStore::init_csr)
+        .def("add", &DeviceStore::add, py::call_guard<py::gil_scoped_release>())
+        .def("init", &DeviceStore::init, py::call_guard<py::gil_scoped_release>())
+        .def("add_csr", &DeviceStore::add_csr, py::call_guard<py::gil_scoped_release>())
+        .def("init_csr", &DeviceStore::init_csr, py::call_guard<py::gil_scoped_release>())
         .def("ipc_handle", &DeviceStore::ipc_handle)
         .def("open_peers", &DeviceStore::open_peers)
-        .def("update", &DeviceStore::update)
-        .def("update_elems", &DeviceStore::update_elems)
+        .def("update", &DeviceStore::update, py::call_guard<py::gil_scoped_release>())
+        .def("update_elems", &DeviceStore::update_elems, py::call_guard<py::gil_scoped_release>())
         .def("reset_counters", &DeviceStore::reset_counters)
         .def("check_strict", &DeviceStore::check_strict)
-        .def("get_range", &DeviceStore::get_range)
-        .def("gather", &DeviceStore::gather)
-        .def("gather_affine", &DeviceStore::gather_affine)
-        .def("gather_csr", &DeviceStore::gather_csr)
-        .def("csr_lens", &DeviceStore::csr_lens)
-        .def("gather_csr_fast", &DeviceStore::gather_csr_fast)
-        .def("scatter_local", &DeviceStore::scatter_local)
+        .def("get_range", &DeviceStore::get_range, py::call_guard<py::gil_scoped_release>())
+        .def("gather", &DeviceStore::gather, py::call_guard<py::gil_scoped_release>())
+        .def("gather_affine", &DeviceStore::gather_affine, py::call_guard<py::gil_scoped_release>())
+        .def("gather_csr", &DeviceStore::gather_csr, py::call_guard<py::gil_scoped_release>())
+        .def("csr_lens", &DeviceStore::csr_lens, py::call_guard<py::gil_scoped_release>())
+        .def("gather_csr_fast", &DeviceStore::gather_csr_fast, py::call_guard<py::gil_scoped_release>())
+        .def("scatter_local", &DeviceStore::scatter_local, py::call_guard<py::gil_scoped_release>())
         .def("local_shard", &DeviceStore::local_shard)
-        .def("epoch_begin", &DeviceStore::epoch_begin)
-        .def("epoch_end", &DeviceStore::epoch_end)
+        .def("epoch_begin", &DeviceStore::epoch_begin, py::call_guard<py::gil_scoped_release>())
+        .def("epoch_end", &DeviceStore::epoch_end, py::call_guard<py::gil_scoped_release>())
         .def("epoch_active", &DeviceStore::epoch_active)
         .def("query", &DeviceStore::query)
         .def("has", &DeviceStore::has)
@@ -68,22 +69,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def(py::init<const std::string&, int, int>(), py::arg("session"),
              py::arg("rank"), py::arg("nparts"))
         .def("shm_name", &HostStore::shm_name)
-        .def("add", &HostStore::add)
-        .def("init", &HostStore::init)
-        .def("add_csr", &HostStore::add_csr)
-        .def("init_csr", &HostStore::init_csr)
+        .def("add", &HostStore::add, py::call_guard<py::gil_scoped_release>())
+        .def("init", &HostStore::init, py::call_guard<py::gil_scoped_release>())
+        .def("add_csr", &HostStore::add_csr, py::call_guard<py::gil_scoped_release>())
+        .def("init_csr", &HostStore::init_csr, py::call_guard<py::gil_scoped_release>())
         .def("open_peers", &HostStore::open_peers)
-        .def("update", &HostStore::update)
-        .def("update_elems", &HostStore::update_elems)
+        .def("update", &HostStore::update, py::call_guard<py::gil_scoped_release>())
+        .def("update_elems", &HostStore::update_elems, py::call_guard<py::gil_scoped_release>())
         .def("reset_counters", &HostStore::reset_counters)
         .def("check_strict", &HostStore::check_strict)
-        .def("get_range", &HostStore::get_range)
-        .def("gather", &HostStore::gather)
-        .def("gather_csr", &HostStore::gather_csr)
-        .def("scatter_local", &HostStore::scatter_local)
+        .def("get_range", &HostStore::get_range, py::call_guard<py::gil_scoped_release>())
+        .def("gather", &HostStore::gather, py::call_guard<py::gil_scoped_release>())
+        .def("gather_csr", &HostStore::gather_csr, py::call_guard<py::gil_scoped_release>())
+        .def("scatter_local", &HostStore::scatter_local, py::call_guard<py::gil_scoped_release>())
         .def("local_shard", &HostStore::local_shard)
-        .def("epoch_begin", &HostStore::epoch_begin)
-        .def("epoch_end", &HostStore::epoch_end)
+        .def("epoch_begin", &HostStore::epoch_begin, py::call_guard<py::gil_scoped_release>())
+        .def("epoch_end", &HostStore::epoch_end, py::call_guard<py::gil_scoped_release>())
         .def("epoch_active", &HostStore::epoch_active)
         .def("query", &HostStore::query)
         .def("has", &HostStore::has)
