@@ -213,3 +213,32 @@ def _w_perm_divergence(rank, world):
 
 def test_perm_divergence_detected_ws2():
     run_dist(_w_perm_divergence, 2)
+
+
+def test_concurrent_gathers_thread_safe(store):
+    """The heavy native ops release the GIL (r2), so several Python threads
+    can gather at once; HostPool serializes its task state internally.
+    Verifies results stay correct under 4 concurrent gather threads."""
+    import threading
+
+    rows, dim = 20000, 16
+    base = torch.arange(rows, dtype=torch.float32).unsqueeze(1).repeat(1, dim)
+    store.add("mt", base)
+    errs = []
+
+    def worker(seed):
+        try:
+            rng = np.random.default_rng(seed)
+            for _ in range(20):
+                idx = torch.from_numpy(rng.integers(0, rows, size=4096))
+                out = store.get_batch("mt", idx)
+                assert torch.equal(out[:, 0], idx.to(torch.float32))
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=worker, args=(s,)) for s in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs, errs
