@@ -19,6 +19,8 @@ gather-kernel -> all_to_all_single -> scatter-kernel.
 """
 from __future__ import annotations
 
+import math
+
 import torch
 
 from .store import DDStore
@@ -128,6 +130,63 @@ def _reshuffle_csr(store: DDStore, name: str, seed: int) -> None:
     store.add_csr(name, values, lens)
 
 
+def cycle_order_device(perm: torch.Tensor):
+    """Concatenated-cycle traversal of a permutation via pointer-doubling
+    list ranking -- O(n log n) work but fully parallel, so it runs on the
+    GPU in ~seconds where the native sequential walk (dependent random
+    loads at DRAM latency) takes ~90 s at 537M rows.
+
+    Produces EXACTLY the same (order, starts) as the serial walk: cycles
+    are emitted by ascending minimum element, which is precisely the
+    discovery order of a walk that scans start candidates in ascending
+    index order; within a cycle the traversal is leader, perm[leader], ...
+
+    Uses int32 working tensors when n < 2^31 (the doubling temps dominate
+    transient memory: ~20 GB at 537M rows, fitting beside a 256 GiB shard).
+    """
+    n = perm.numel()
+    dev = perm.device
+    if n == 0:
+        z = torch.zeros(1, dtype=torch.int64)
+        return torch.empty(0, dtype=torch.int64, device=dev), z
+    idt = torch.int32 if n < 2**31 else torch.int64
+    rounds = max(1, int(math.ceil(math.log2(max(n, 2)))))
+
+    # leader[e] = min element of e's cycle (min-propagation doubling)
+    leader = torch.arange(n, dtype=idt, device=dev)
+    jump = perm.to(idt)
+    for _ in range(rounds):
+        leader = torch.minimum(leader, leader[jump.long()])
+        jump = jump[jump.long()]
+    del jump
+
+    # distance to the cycle's LAST traversal element (the one whose
+    # successor is the leader), via doubling on the leader-broken chain
+    succ = perm.to(idt)
+    is_last = succ.long() == leader.long()
+    d = torch.ones(n, dtype=idt, device=dev)
+    d[is_last] = 0
+    succ = torch.where(is_last, torch.arange(n, dtype=idt, device=dev), succ)
+    for _ in range(rounds):
+        sl = succ.long()
+        d = d + d[sl]
+        succ = succ[sl]
+    del succ, is_last
+
+    # position within cycle = (len-1) - dist_to_last = d[leader] - d
+    ll = leader.long()
+    pos = (d[ll] - d).long()
+    del d
+    # cycles ordered by ascending leader; bases from their sizes
+    uleaders, counts = torch.unique(ll, return_counts=True)
+    starts = torch.zeros(uleaders.numel() + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=starts[1:])
+    cyc = torch.searchsorted(uleaders, ll)
+    order = torch.empty(n, dtype=torch.int64, device=dev)
+    order[starts[cyc] + pos] = torch.arange(n, dtype=torch.int64, device=dev)
+    return order, starts.cpu()
+
+
 def reshuffle_epoch_chunked(
     store: DDStore, name: str, seed: int, max_chunk_bytes: int = 1 << 30
 ) -> None:
@@ -174,12 +233,18 @@ def reshuffle_epoch_chunked(
     g.manual_seed(int(seed))
     perm = torch.randperm(ntotal, generator=g, device=dev)
     _check_perm_agreement(store, perm)
-    order, starts = _C.cycle_order(perm.cpu())
+    if store.mode == "hip":
+        # parallel list ranking on the GPU (~2 s at 537M rows); the serial
+        # host walk is the CPU-mode fallback (~90 s there)
+        order, starts = cycle_order_device(perm)
+    else:
+        order, starts = _C.cycle_order(perm.cpu())
     del perm
 
-    closing_pos = starts[1:] - 1  # traversal positions that close a cycle
+    odev = order.device  # device-resident on GPU stores (parallel ranking)
+    closing_pos = (starts[1:] - 1).to(odev)  # positions that close a cycle
     closing_slot = order[closing_pos]
-    heads = order[starts[:-1]]  # cycle head slot, aligned with closing_slot
+    heads = order[starts[:-1].to(odev)]  # cycle head slot, aligned
     # save head rows for the closing writes THIS rank owns, before any write
     close_mine = (closing_slot >= p0) & (closing_slot < p1)
     my_heads = heads[close_mine]
@@ -207,9 +272,9 @@ def reshuffle_epoch_chunked(
             w_mine = W[mine]
         if w_mine.numel():
             pos = (
-                torch.arange(a, b, dtype=torch.int64)
+                torch.arange(a, b, dtype=torch.int64, device=odev)
                 if owns_all
-                else torch.arange(a, b, dtype=torch.int64)[mine]
+                else torch.arange(a, b, dtype=torch.int64, device=odev)[mine]
             )
             is_close = torch.isin(pos, closing_pos)
             src = order[torch.clamp(pos + 1, max=ntotal - 1)]

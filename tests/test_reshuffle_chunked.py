@@ -115,3 +115,23 @@ def test_chunked_rejects_csr():
     with pytest.raises(ValueError, match="fixed-stride"):
         reshuffle_epoch_chunked(s, "c", 1)
     s.free()
+
+
+@pytest.mark.parametrize("n", [1, 2, 5, 64, 257, 5000])
+def test_cycle_order_device_matches_native(n):
+    """The pointer-doubling traversal must equal the serial walk EXACTLY
+    (same cycle order, same within-cycle order) -- torch ops run this on
+    CPU here; on GPU the same code path feeds the capacity reshuffle."""
+    from ddstore_amd.reshuffle import cycle_order_device
+
+    for seed in range(3):
+        perm = torch.randperm(n, generator=torch.Generator().manual_seed(seed))
+        o1, s1 = _C.cycle_order(perm)
+        o2, s2 = cycle_order_device(perm)
+        assert torch.equal(o1, o2.cpu()), (n, seed)
+        assert torch.equal(s1, s2.cpu()), (n, seed)
+    # identity and reversal corner cases
+    for perm in [torch.arange(n), torch.arange(n - 1, -1, -1)]:
+        o1, s1 = _C.cycle_order(perm.contiguous())
+        o2, s2 = cycle_order_device(perm.contiguous())
+        assert torch.equal(o1, o2.cpu()) and torch.equal(s1, s2.cpu())
