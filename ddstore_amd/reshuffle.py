@@ -173,17 +173,24 @@ def cycle_order_device(perm: torch.Tensor):
         succ = succ[sl]
     del succ, is_last
 
-    # position within cycle = (len-1) - dist_to_last = d[leader] - d
-    ll = leader.long()
-    pos = (d[ll] - d).long()
-    del d
-    # cycles ordered by ascending leader; bases from their sizes
-    uleaders, counts = torch.unique(ll, return_counts=True)
+    # cycle heads = elements that are their own leader (the cycle minimum);
+    # lengths come free from the ranking (d[head] = len-1) -- this avoids a
+    # torch.unique whose sort buffers OOM'd beside a 256 GiB shard
+    head_mask = leader == torch.arange(n, dtype=idt, device=dev)
+    uleaders = torch.nonzero(head_mask, as_tuple=True)[0]  # sorted asc, tiny
+    del head_mask
+    counts = (d[uleaders] + 1).long()
     starts = torch.zeros(uleaders.numel() + 1, dtype=torch.int64, device=dev)
     torch.cumsum(counts, 0, out=starts[1:])
-    cyc = torch.searchsorted(uleaders, ll)
+    # position within cycle = (len-1) - dist_to_last = d[leader] - d
+    pos = d[leader.long()] - d  # narrow dtype; transient i64 index only
+    del d
+    cyc = torch.searchsorted(uleaders.to(idt), leader,
+                             out_int32=(idt == torch.int32))
+    del leader
     order = torch.empty(n, dtype=torch.int64, device=dev)
-    order[starts[cyc] + pos] = torch.arange(n, dtype=torch.int64, device=dev)
+    order[starts[cyc.long()] + pos.long()] = torch.arange(
+        n, dtype=torch.int64, device=dev)
     return order, starts.cpu()
 
 
