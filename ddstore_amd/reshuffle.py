@@ -144,6 +144,11 @@ def cycle_order_device(perm: torch.Tensor):
     Uses int32 working tensors when n < 2^31 (the doubling temps dominate
     transient memory: ~20 GB at 537M rows, fitting beside a 256 GiB shard).
     """
+    if isinstance(perm, list):  # ownership transfer: we may free it early
+        (perm,) = perm
+        owned = True
+    else:
+        owned = False
     n = perm.numel()
     dev = perm.device
     if n == 0:
@@ -151,6 +156,7 @@ def cycle_order_device(perm: torch.Tensor):
         return torch.empty(0, dtype=torch.int64, device=dev), z
     idt = torch.int32 if n < 2**31 else torch.int64
     rounds = max(1, int(math.ceil(math.log2(max(n, 2)))))
+    chunk = 1 << 26  # bound the i64 index temps beside near-capacity shards
 
     # leader[e] = min element of e's cycle (min-propagation doubling)
     leader = torch.arange(n, dtype=idt, device=dev)
@@ -163,15 +169,18 @@ def cycle_order_device(perm: torch.Tensor):
     # distance to the cycle's LAST traversal element (the one whose
     # successor is the leader), via doubling on the leader-broken chain
     succ = perm.to(idt)
+    if owned:
+        del perm  # free 8n bytes before the ranking peak
     is_last = succ.long() == leader.long()
     d = torch.ones(n, dtype=idt, device=dev)
     d[is_last] = 0
     succ = torch.where(is_last, torch.arange(n, dtype=idt, device=dev), succ)
+    del is_last
     for _ in range(rounds):
         sl = succ.long()
         d = d + d[sl]
         succ = succ[sl]
-    del succ, is_last
+    del succ, sl
 
     # cycle heads = elements that are their own leader (the cycle minimum);
     # lengths come free from the ranking (d[head] = len-1) -- this avoids a
@@ -182,15 +191,23 @@ def cycle_order_device(perm: torch.Tensor):
     counts = (d[uleaders] + 1).long()
     starts = torch.zeros(uleaders.numel() + 1, dtype=torch.int64, device=dev)
     torch.cumsum(counts, 0, out=starts[1:])
-    # position within cycle = (len-1) - dist_to_last = d[leader] - d
-    pos = d[leader.long()] - d  # narrow dtype; transient i64 index only
-    del d
-    cyc = torch.searchsorted(uleaders.to(idt), leader,
-                             out_int32=(idt == torch.int32))
-    del leader
+    # position within cycle = (len-1) - dist_to_last = d[leader] - d, and
+    # slot = starts[cycle(e)] + pos(e); both computed in bounded chunks so
+    # the unavoidable i64 temps stay ~0.5 GB each (a full-width pass OOM'd
+    # beside the 256 GiB shard)
+    ul = uleaders.to(idt)
     order = torch.empty(n, dtype=torch.int64, device=dev)
-    order[starts[cyc.long()] + pos.long()] = torch.arange(
-        n, dtype=torch.int64, device=dev)
+    ar = None
+    for lo in range(0, n, chunk):
+        hi = min(lo + chunk, n)
+        lead_c = leader[lo:hi].long()
+        pos_c = (d[lead_c] - d[lo:hi]).long()
+        cyc_c = torch.searchsorted(ul, leader[lo:hi]).long()
+        if ar is None or ar.numel() != hi - lo:
+            ar = torch.arange(lo, hi, dtype=torch.int64, device=dev)
+        else:
+            ar = torch.arange(lo, hi, dtype=torch.int64, device=dev)
+        order[starts[cyc_c] + pos_c] = ar
     return order, starts.cpu()
 
 
@@ -242,11 +259,20 @@ def reshuffle_epoch_chunked(
     _check_perm_agreement(store, perm)
     if store.mode == "hip":
         # parallel list ranking on the GPU (~2 s at 537M rows); the serial
-        # host walk is the CPU-mode fallback (~90 s there)
-        order, starts = cycle_order_device(perm)
+        # host walk is the CPU-mode fallback (~90 s there). Ownership of
+        # perm transfers so its 8n bytes free before the ranking peak.
+        box = [perm]
+        perm_cpu = perm.cpu()  # fallback copy (host RAM)
+        del perm
+        try:
+            order, starts = cycle_order_device(box)
+        except torch.OutOfMemoryError:
+            # ranking temps did not fit beside the shard: serial host walk
+            order, starts = _C.cycle_order(perm_cpu)
+        del perm_cpu
     else:
         order, starts = _C.cycle_order(perm.cpu())
-    del perm
+        del perm
 
     odev = order.device  # device-resident on GPU stores (parallel ranking)
     closing_pos = (starts[1:] - 1).to(odev)  # positions that close a cycle
