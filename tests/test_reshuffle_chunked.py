@@ -135,3 +135,17 @@ def test_cycle_order_device_matches_native(n):
         o1, s1 = _C.cycle_order(perm.contiguous())
         o2, s2 = cycle_order_device(perm.contiguous())
         assert torch.equal(o1, o2.cpu()) and torch.equal(s1, s2.cpu())
+
+
+def test_cycle_order_device_property():
+    """Hypothesis-style randomized check across many shapes: the parallel
+    ranking equals the serial walk for every permutation tried."""
+    from ddstore_amd.reshuffle import cycle_order_device
+
+    g = torch.Generator().manual_seed(2024)
+    for _ in range(60):
+        n = int(torch.randint(1, 3000, (1,), generator=g))
+        perm = torch.randperm(n, generator=g)
+        o1, s1 = _C.cycle_order(perm)
+        o2, s2 = cycle_order_device(perm)
+        assert torch.equal(o1, o2) and torch.equal(s1, s2), n
